@@ -1,0 +1,174 @@
+"""Core domain tests (mirrors pkg/core/*_test.go: spec round-trips,
+allocation sizing, transition penalties, saturation, power curve)."""
+
+import math
+
+import pytest
+
+from wva_amd.config import ServerLoadSpec
+from wva_amd.core import Accelerator, Allocation, System, create_allocation
+from fixtures import MI355X, make_system, server_spec
+
+
+class TestAccelerator:
+    def test_power_curve(self):
+        acc = Accelerator(MI355X)
+        acc.calculate()
+        assert acc.power(0.0) == pytest.approx(140.0)
+        assert acc.power(0.6) == pytest.approx(900.0)
+        assert acc.power(1.0) == pytest.approx(1400.0)
+        # piecewise linear midpoints
+        assert acc.power(0.3) == pytest.approx(140 + (900 - 140) / 0.6 * 0.3)
+        assert acc.power(0.8) == pytest.approx(900 + (1400 - 900) / 0.4 * 0.2)
+
+
+class TestServiceClass:
+    def test_priority_clamping(self):
+        system, _ = make_system()
+        system.add_service_class("weird", 1000)
+        assert system.service_class("weird").priority == 100
+        system.add_service_class("neg", 0)
+        assert system.service_class("neg").priority == 100
+
+
+class TestCreateAllocation:
+    def test_feasible_allocation(self):
+        system, _ = make_system()
+        system.calculate()
+        alloc = create_allocation(system, "s1:default", "MI355X")
+        assert alloc is not None
+        assert alloc.accelerator == "MI355X"
+        assert alloc.num_replicas >= 1
+        assert alloc.cost == pytest.approx(85.0 * alloc.num_replicas)
+        assert alloc.itl <= 20.0 * 1.001  # meets Premium SLO
+        assert alloc.max_arrv_rate_per_replica > 0
+
+    def test_unknown_names_return_none(self):
+        system, _ = make_system()
+        assert create_allocation(system, "nope", "MI355X") is None
+        assert create_allocation(system, "s1:default", "nope") is None
+
+    def test_infeasible_slo_returns_none(self):
+        # L40S alpha=20 > Premium ITL target of 20*... below alpha+beta
+        system, _ = make_system()
+        alloc = create_allocation(system, "s1:default", "L40S")
+        assert alloc is None
+
+    def test_replica_scaling_with_load(self):
+        sys_lo, _ = make_system(servers=[server_spec("s:ns", arrival_rate=60.0)])
+        sys_hi, _ = make_system(servers=[server_spec("s:ns", arrival_rate=6000.0)])
+        lo = create_allocation(sys_lo, "s:ns", "MI355X")
+        hi = create_allocation(sys_hi, "s:ns", "MI355X")
+        assert hi.num_replicas > lo.num_replicas
+        # replica count = ceil(totalRate / rate*)
+        total_rate = 6000.0 / 60.0
+        expected = max(
+            math.ceil(total_rate / (hi.max_arrv_rate_per_replica * 1000.0)), 1
+        )
+        assert hi.num_replicas == expected
+
+    def test_zero_load_min_replicas(self):
+        system, _ = make_system(servers=[server_spec("s:ns", arrival_rate=0.0, min_replicas=1)])
+        alloc = create_allocation(system, "s:ns", "MI355X")
+        assert alloc.num_replicas == 1
+        assert alloc.rho == 0.0
+        assert alloc.cost == pytest.approx(85.0)
+
+    def test_zero_load_scale_to_zero(self):
+        system, _ = make_system(servers=[server_spec("s:ns", arrival_rate=0.0, min_replicas=0)])
+        alloc = create_allocation(system, "s:ns", "MI355X")
+        assert alloc.num_replicas == 0
+        assert alloc.accelerator == ""
+        assert alloc.cost == 0.0
+
+    def test_acc_count_multiplies_cost(self):
+        system, _ = make_system(
+            servers=[server_spec("s:ns", model="llama-70b", arrival_rate=600.0, max_batch=8)]
+        )
+        alloc = create_allocation(system, "s:ns", "MI355X")
+        assert alloc is not None
+        # llama-70b on MI355X uses acc_count=4 instances per replica
+        assert alloc.cost == pytest.approx(85.0 * 4 * alloc.num_replicas)
+
+
+class TestTransitionPenalty:
+    def test_same_acc_same_replicas(self):
+        a = Allocation(accelerator="MI355X", num_replicas=2, cost=170.0)
+        b = Allocation(accelerator="MI355X", num_replicas=2, cost=170.0)
+        assert a.transition_penalty(b) == 0.0
+
+    def test_same_acc_scaling(self):
+        a = Allocation(accelerator="MI355X", num_replicas=2, cost=170.0)
+        b = Allocation(accelerator="MI355X", num_replicas=3, cost=255.0)
+        assert a.transition_penalty(b) == pytest.approx(85.0)
+
+    def test_acc_change_penalized(self):
+        a = Allocation(accelerator="MI355X", num_replicas=2, cost=170.0)
+        b = Allocation(accelerator="MI300X", num_replicas=2, cost=130.0)
+        assert a.transition_penalty(b) == pytest.approx(0.1 * (170 + 130) + (130 - 170))
+
+
+class TestSaturation:
+    def test_saturated(self):
+        a = Allocation(accelerator="MI355X", num_replicas=1, max_arrv_rate_per_replica=0.001)
+        # maxRPM = 0.001 * 1000 * 60 = 60 req/min
+        assert not a.saturated(59.0)
+        assert a.saturated(61.0)
+
+
+class TestServer:
+    def test_keep_accelerator_pins_candidates(self):
+        system, _ = make_system(
+            servers=[server_spec("s:ns", keep_accelerator=True, cur_accelerator="MI300X", cur_replicas=1)]
+        )
+        system.calculate()
+        server = system.server("s:ns")
+        assert set(server.all_allocations) == {"MI300X"}
+
+    def test_no_pin_enumerates_all_feasible(self):
+        system, _ = make_system()
+        system.calculate()
+        server = system.server("s1:default")
+        # L40S infeasible under Premium SLO; MI355X and MI300X feasible
+        assert set(server.all_allocations) == {"MI355X", "MI300X"}
+
+    def test_value_is_transition_penalty(self):
+        system, _ = make_system(
+            servers=[server_spec("s:ns", cur_accelerator="MI355X", cur_replicas=1)]
+        )
+        system.calculate()
+        server = system.server("s:ns")
+        alloc = server.all_allocations["MI355X"]
+        cur = server.cur_allocation
+        assert alloc.value == pytest.approx(cur.transition_penalty(alloc))
+
+
+class TestSystem:
+    def test_allocate_by_type_and_solution(self):
+        system, _ = make_system(
+            servers=[
+                server_spec("a:ns", arrival_rate=600.0),
+                server_spec("b:ns", model="llama-70b", arrival_rate=600.0),
+            ]
+        )
+        system.calculate()
+        for server in system.servers.values():
+            server.set_allocation(server.all_allocations["MI355X"])
+        system.allocate_by_type()
+        t = system.allocation_by_type["AMD-MI355X-288GB"]
+        a = system.server("a:ns").allocation
+        b = system.server("b:ns").allocation
+        assert t.count == a.num_replicas * 1 + b.num_replicas * 4
+        assert t.cost == pytest.approx(a.cost + b.cost)
+
+        sol = system.generate_solution()
+        assert set(sol.spec) == {"a:ns", "b:ns"}
+        assert sol.spec["a:ns"].accelerator == "MI355X"
+        assert sol.spec["a:ns"].load.arrival_rate == 600.0
+
+    def test_registry_removal(self):
+        system, _ = make_system()
+        system.remove_accelerator("L40S")
+        assert system.accelerator("L40S") is None
+        with pytest.raises(KeyError):
+            system.remove_accelerator("L40S")
