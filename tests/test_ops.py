@@ -1,0 +1,135 @@
+"""Native batched-solver tests: parity of the C++/HIP paths against the
+pure-Python analyzer (the semantic reference)."""
+
+import numpy as np
+import pytest
+
+from wva_amd.ops import BatchedAllocationSolver, native_available, solve_problems
+from wva_amd.ops.batched import (
+    PROBLEM_FIELDS,
+    R_FEASIBLE,
+    R_ITL,
+    R_RATE_STAR,
+    R_REPLICAS,
+    R_RHO,
+    R_TTFT,
+    _solve_problems_python,
+)
+from fixtures import make_system, server_spec
+
+RNG = np.random.default_rng(7)
+
+
+def random_problems(n, max_batch_hi=256):
+    rows = []
+    for _ in range(n):
+        alpha = RNG.uniform(2.0, 25.0)
+        beta = RNG.uniform(0.005, 0.5)
+        gamma = RNG.uniform(2.0, 60.0)
+        delta = RNG.uniform(0.001, 0.2)
+        in_tok = float(RNG.integers(0, 2048))
+        out_tok = float(RNG.integers(1, 1024))
+        n_batch = float(RNG.integers(1, max_batch_hi))
+        # targets comfortably above alpha/gamma so most problems are feasible
+        t_itl = alpha + RNG.uniform(0.5, 30.0)
+        t_ttft = gamma + delta * in_tok * 2 + RNG.uniform(10.0, 5000.0)
+        t_tps = 0.0 if RNG.random() < 0.7 else RNG.uniform(100.0, 5000.0)
+        total_rate = RNG.uniform(0.01, 500.0)
+        min_rep = float(RNG.integers(0, 3))
+        rows.append(
+            [alpha, beta, gamma, delta, in_tok, out_tok, n_batch, t_ttft, t_itl, t_tps, total_rate, min_rep]
+        )
+    return np.asarray(rows, dtype=np.float64)
+
+
+def assert_results_close(a, b, rtol=1e-6):
+    assert a.shape == b.shape
+    np.testing.assert_array_equal(a[:, R_FEASIBLE], b[:, R_FEASIBLE])
+    feas = a[:, R_FEASIBLE] == 1.0
+    np.testing.assert_array_equal(a[feas, R_REPLICAS], b[feas, R_REPLICAS])
+    for col in (R_RATE_STAR, R_ITL, R_TTFT, R_RHO):
+        np.testing.assert_allclose(a[feas, col], b[feas, col], rtol=rtol, atol=1e-9)
+
+
+@pytest.mark.skipif(not native_available(), reason="native extension not built")
+class TestNativeCPUParity:
+    def test_random_problem_parity(self):
+        problems = random_problems(64)
+        got = solve_problems(problems, device="cpu")
+        want = _solve_problems_python(problems)
+        # bisection iterates on floats: tiny tolerance differences between
+        # numpy-vectorized and scalar C++ sums can shift the found lambda
+        assert_results_close(got, want, rtol=1e-4)
+        assert got[:, R_FEASIBLE].sum() > 0  # exercise the feasible path
+
+    def test_infeasible_targets(self):
+        problems = random_problems(4)
+        problems[:, 8] = 0.01  # ITL target below alpha: infeasible
+        got = solve_problems(problems, device="cpu")
+        assert (got[:, R_FEASIBLE] == 0.0).all()
+
+    def test_empty_batch(self):
+        out = solve_problems(np.zeros((0, PROBLEM_FIELDS)))
+        assert out.shape == (0, 6)
+
+
+class TestBatchedSystemParity:
+    def test_matches_scalar_system_calculate(self):
+        servers = [
+            server_spec("a:ns", arrival_rate=600.0),
+            server_spec("b:ns", model="llama-70b", arrival_rate=1200.0, max_batch=16),
+            server_spec("c:ns", class_name="Freemium", arrival_rate=90.0),
+            server_spec("z:ns", arrival_rate=0.0),  # zero-load path
+            server_spec("k:ns", keep_accelerator=True, cur_accelerator="MI300X", cur_replicas=2),
+        ]
+        sys_scalar, _ = make_system(servers=servers)
+        sys_batch, _ = make_system(servers=servers)
+        sys_scalar.calculate()
+        BatchedAllocationSolver().calculate(sys_batch)
+        for name in sys_scalar.servers:
+            sa = sys_scalar.server(name).all_allocations
+            ba = sys_batch.server(name).all_allocations
+            assert set(sa) == set(ba), name
+            for acc in sa:
+                x, y = sa[acc], ba[acc]
+                assert x.num_replicas == y.num_replicas
+                assert x.batch_size == y.batch_size
+                np.testing.assert_allclose(x.cost, y.cost, rtol=1e-6)
+                np.testing.assert_allclose(x.value, y.value, rtol=1e-6, atol=1e-9)
+                np.testing.assert_allclose(x.itl, y.itl, rtol=1e-4)
+                np.testing.assert_allclose(x.ttft, y.ttft, rtol=1e-4, atol=1e-6)
+                np.testing.assert_allclose(
+                    x.max_arrv_rate_per_replica, y.max_arrv_rate_per_replica, rtol=1e-4
+                )
+
+
+@pytest.mark.gpu
+class TestGPU:
+    def test_native_loaded_and_has_hip(self):
+        import torch
+
+        assert torch.cuda.is_available()
+        from wva_amd.ops import get_native
+
+        native = get_native()
+        assert native is not None and native.HAS_HIP
+
+    def test_gpu_matches_cpu(self):
+        problems = random_problems(256)
+        cpu = solve_problems(problems, device="cpu")
+        gpu = solve_problems(problems, device="cuda")
+        assert_results_close(gpu, cpu, rtol=1e-4)
+        assert cpu[:, R_FEASIBLE].sum() > 100
+
+    def test_gpu_matches_python_reference(self):
+        problems = random_problems(32)
+        gpu = solve_problems(problems, device="cuda")
+        ref = _solve_problems_python(problems)
+        assert_results_close(gpu, ref, rtol=1e-4)
+
+    def test_large_batch_limit_falls_back(self):
+        problems = random_problems(8)
+        problems[0, 6] = 1024.0  # beyond the LDS-resident limit
+        out = solve_problems(problems, device="cuda")
+        ref = solve_problems(problems, device="cpu")
+        assert_results_close(out, ref, rtol=1e-4)

@@ -1,0 +1,157 @@
+// Shared scalar math for the batched state-dependent M/M/1/K allocation
+// sizing kernel.  Mirrors wva_amd/analyzer (which itself re-implements the
+// behavior of /root/reference/pkg/analyzer) in double precision:
+//   - service rates  mu(b) = b / (prefill(b) + numDecode * decode(b))
+//   - product-form state probabilities in log space (softmax-normalized)
+//   - monotone bisection with boundary-region classification
+//   - Size(): max rate meeting TTFT/ITL targets; TPS as lambdaMax*0.9
+//
+// Used by both the CPU (OpenMP) path and the gfx950 HIP kernel.
+#pragma once
+
+#include <cmath>
+
+#if defined(__HIPCC__)
+#define WVA_HD __host__ __device__ inline
+#else
+#define WVA_HD inline
+#endif
+
+namespace wva {
+
+constexpr double kEpsilon = 0.001;           // rate-range disturbance
+constexpr double kStabilityFraction = 0.1;   // TPS backoff fraction
+constexpr double kTolerance = 1e-6;          // bisection relative tolerance
+constexpr int kMaxIterations = 100;          // bisection iteration cap
+constexpr int kMaxQueueToBatchRatio = 10;    // K = (1 + ratio) * N
+
+// Problem layout: one row of 12 doubles.
+enum ProblemField {
+  P_ALPHA = 0,    // decode base (ms)
+  P_BETA,         // decode slope (ms per batch unit)
+  P_GAMMA,        // prefill base (ms)
+  P_DELTA,        // prefill slope (ms per token per batch unit)
+  P_IN_TOKENS,    // average input tokens
+  P_OUT_TOKENS,   // average output tokens (K >= 1)
+  P_MAX_BATCH,    // N >= 1
+  P_TARGET_TTFT,  // ms; 0 disables
+  P_TARGET_ITL,   // ms; 0 disables
+  P_TARGET_TPS,   // tokens/s; 0 disables
+  P_TOTAL_RATE,   // req/s (> 0; zero-load handled by the host)
+  P_MIN_REPLICAS, // minimum replica count
+  PROBLEM_FIELDS
+};
+
+// Result layout: one row of 6 doubles.
+enum ResultField {
+  R_FEASIBLE = 0, // 1.0 feasible / 0.0 infeasible
+  R_REPLICAS,     // number of replicas
+  R_RATE_STAR,    // max per-replica rate meeting targets (req/s)
+  R_ITL,          // predicted decode time per token at the final rate (ms)
+  R_TTFT,         // predicted wait + prefill at the final rate (ms)
+  R_RHO,          // avg-in-service / max-batch, clamped to [0,1]
+  RESULT_FIELDS
+};
+
+struct Parms {
+  double alpha, beta, gamma, delta;
+  double in_tokens;
+  int out_tokens;
+  int max_batch;   // N
+  int num_decode;  // out_tokens-1, or 1 for decode-only single-token
+};
+
+WVA_HD double prefill_time(const Parms &p, double batch) {
+  return p.in_tokens == 0.0 ? 0.0 : p.gamma + p.delta * p.in_tokens * batch;
+}
+
+WVA_HD double decode_time(const Parms &p, double batch) {
+  return p.alpha + p.beta * batch;
+}
+
+// state-dependent service rate for batch size b in [1, N] (req/ms)
+WVA_HD double serv_rate(const Parms &p, int b) {
+  return (double)b / (prefill_time(p, (double)b) + p.num_decode * decode_time(p, (double)b));
+}
+
+// log(mu(state n)) for the birth-death recursion, n in [0, K-1]
+WVA_HD double log_mu(const Parms &p, int n) {
+  int b = n + 1;
+  if (b > p.max_batch) b = p.max_batch;
+  return log(serv_rate(p, b));
+}
+
+WVA_HD bool within_tolerance(double x, double value, double tolerance) {
+  if (x == value) return true;
+  if (value == 0.0 || tolerance < 0.0) return false;
+  return fabs((x - value) / value) <= tolerance;
+}
+
+// effective concurrency: solve the service-time identity for n, clamped
+WVA_HD double effective_concurrency(const Parms &p, double avg_serv_time) {
+  double tokens = (double)(p.out_tokens - 1);
+  double numer = avg_serv_time - (p.gamma + p.alpha * tokens);
+  double denom = p.delta * p.in_tokens + p.beta * tokens;
+  if (denom == 0.0) return numer > 0.0 ? (double)p.max_batch : 0.0;
+  double n = numer / denom;
+  if (n < 0.0) n = 0.0;
+  if (n > (double)p.max_batch) n = (double)p.max_batch;
+  return n;
+}
+
+struct Stats {
+  double throughput;  // req/ms
+  double wait;        // ms
+  double serv;        // ms
+  double n_serv;      // avg requests in service
+};
+
+// Scalar model evaluation at arrival rate lam (req/ms) given the inclusive
+// cumulative sum cum[n] = sum_{i<=n} log_mu(i), n in [0, K-1].
+WVA_HD Stats eval_model(const Parms &p, const double *cum, int K, double lam) {
+  double loglam = log(lam);
+  // max of logp over n = 0..K (logp(0) = 0)
+  double m = 0.0;
+  for (int n = 1; n <= K; ++n) {
+    double lp = n * loglam - cum[n - 1];
+    if (lp > m) m = lp;
+  }
+  int num = p.max_batch;  // serv_rate array length
+  double S = 0.0, Ni = 0.0, Snum = 0.0, Ninum = 0.0, eK = 0.0;
+  for (int n = 0; n <= K; ++n) {
+    double lp = (n == 0) ? 0.0 : n * loglam - cum[n - 1];
+    double e = exp(lp - m);
+    S += e;
+    Ni += n * e;
+    if (n <= num) {
+      Snum += e;
+      Ninum += n * e;
+    }
+    if (n == K) eK = e;
+  }
+  Stats st;
+  st.throughput = lam * (1.0 - eK / S);
+  double n_sys = Ni / S;
+  st.n_serv = Ninum / S + (1.0 - Snum / S) * num;
+  if (st.throughput == 0.0) {
+    st.wait = st.serv = 0.0;
+  } else {
+    double resp = n_sys / st.throughput;
+    st.serv = st.n_serv / st.throughput;
+    st.wait = resp - st.serv;
+    if (st.wait < 0.0) st.wait = 0.0;
+  }
+  return st;
+}
+
+WVA_HD double eval_ttft_of(const Parms &p, const Stats &st) {
+  double effc = effective_concurrency(p, st.serv);
+  return st.wait + prefill_time(p, effc);
+}
+
+WVA_HD double eval_itl_of(const Parms &p, const Stats &st) {
+  double effc = effective_concurrency(p, st.serv);
+  return decode_time(p, effc);
+}
+
+}  // namespace wva

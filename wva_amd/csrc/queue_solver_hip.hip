@@ -1,0 +1,267 @@
+// gfx950 (CDNA4) batched allocation-sizing kernel.
+//
+// One 256-thread workgroup (4 wave64) per problem.  The whole analytic
+// pipeline of core.create_allocation (service-rate table, log-space
+// product-form M/M/1/K probabilities, TTFT/ITL bisection, replica count,
+// final per-replica analysis) runs in-workgroup:
+//
+//   - the cumulative log-service-rate table (K = 11*N doubles) lives in LDS
+//     and is built with a chunked parallel scan — every later model
+//     evaluation is LDS-bandwidth bound, never HBM;
+//   - each model evaluation is two strided passes over the K+1 states with
+//     wave64 __shfl_down reductions (max, then 4 sums + boundary term);
+//     consecutive lanes touch consecutive doubles -> conflict-free
+//     ds_read_b64 (bank = (a/4) % 64);
+//   - the bisection control flow is uniform across the workgroup (all
+//     decisions derive from broadcast reduction results), so the
+//     __syncthreads() inside evaluations are safe;
+//   - LDS budget: (K + 256 + 32) doubles <= 64 KiB for N <= ~700 (the
+//     binding routes larger batch limits to the CPU path).  At N = 256
+//     (the collector's default max batch) that is ~25 KiB -> 2+ workgroups
+//     per CU, and a fleet-sized batch (hundreds of (server, accelerator)
+//     pairs) fills all 256 CUs.
+//
+// Numerics are double throughout, matching the Python analyzer; math
+// parity is covered by tests/test_ops.py (CPU) and the @gpu numerics
+// tests.
+
+#include <hip/hip_runtime.h>
+
+#include "queue_core.h"
+
+namespace wva {
+
+#define WVA_THREADS 256
+#define WVA_WAVES (WVA_THREADS / 64)
+
+// Reduction scratch layout (doubles, after cum[max_k] in dynamic LDS):
+//   red[0..WVA_WAVES*4-1]  per-wave partial sums (4 values at once)
+//   red[16]                boundary term exp(logp(K) - m)
+//   red[17..17+WVA_WAVES]  per-wave max partials
+
+struct WgEval {
+  const Parms &p;
+  const double *cum;  // LDS, K entries
+  double *red;        // LDS scratch
+  int K;
+
+  __device__ Stats eval(double lam) const {
+    const int tid = threadIdx.x;
+    const int wave = tid >> 6;
+    const int lane = tid & 63;
+    const double loglam = log(lam);
+
+    // pass 1: max of logp over n = 0..K (logp(0) = 0 handled by init)
+    double lmax = (tid == 0) ? 0.0 : -INFINITY;
+    for (int n = tid + 1; n <= K; n += WVA_THREADS) {
+      double lp = (double)n * loglam - cum[n - 1];
+      lmax = fmax(lmax, lp);
+    }
+    for (int off = 32; off > 0; off >>= 1)
+      lmax = fmax(lmax, __shfl_down(lmax, off, 64));
+    __syncthreads();  // red may still be read from a previous eval
+    if (lane == 0) red[17 + wave] = lmax;
+    __syncthreads();
+    double m = red[17];
+    for (int w = 1; w < WVA_WAVES; ++w) m = fmax(m, red[17 + w]);
+
+    // pass 2: normalization and moment sums
+    const int num = p.max_batch;
+    double S = 0.0, Ni = 0.0, Snum = 0.0, Ninum = 0.0;
+    if (tid == 0) {
+      double e0 = exp(0.0 - m);  // state 0
+      S += e0;
+      Snum += e0;
+      if (threadIdx.x == 0 && 0 == K) red[16] = e0;
+    }
+    for (int n = tid + 1; n <= K; n += WVA_THREADS) {
+      double lp = (double)n * loglam - cum[n - 1];
+      double e = exp(lp - m);
+      S += e;
+      Ni += (double)n * e;
+      if (n <= num) {
+        Snum += e;
+        Ninum += (double)n * e;
+      }
+      if (n == K) red[16] = e;  // exactly one thread owns state K
+    }
+    for (int off = 32; off > 0; off >>= 1) {
+      S += __shfl_down(S, off, 64);
+      Ni += __shfl_down(Ni, off, 64);
+      Snum += __shfl_down(Snum, off, 64);
+      Ninum += __shfl_down(Ninum, off, 64);
+    }
+    if (lane == 0) {
+      red[wave * 4 + 0] = S;
+      red[wave * 4 + 1] = Ni;
+      red[wave * 4 + 2] = Snum;
+      red[wave * 4 + 3] = Ninum;
+    }
+    __syncthreads();
+    S = Ni = Snum = Ninum = 0.0;
+    for (int w = 0; w < WVA_WAVES; ++w) {
+      S += red[w * 4 + 0];
+      Ni += red[w * 4 + 1];
+      Snum += red[w * 4 + 2];
+      Ninum += red[w * 4 + 3];
+    }
+    const double eK = red[16];
+
+    Stats st;
+    st.throughput = lam * (1.0 - eK / S);
+    double n_sys = Ni / S;
+    st.n_serv = Ninum / S + (1.0 - Snum / S) * (double)num;
+    if (st.throughput == 0.0) {
+      st.wait = st.serv = 0.0;
+    } else {
+      double resp = n_sys / st.throughput;
+      st.serv = st.n_serv / st.throughput;
+      st.wait = resp - st.serv;
+      if (st.wait < 0.0) st.wait = 0.0;
+    }
+    return st;
+  }
+
+  __device__ double eval_ttft(double lam) const { return eval_ttft_of(p, eval(lam)); }
+  __device__ double eval_itl(double lam) const { return eval_itl_of(p, eval(lam)); }
+};
+
+// Uniform-control-flow bisection (all threads run it in lockstep; the eval
+// results are identical on every thread because they come from broadcast
+// reductions).  Semantics mirror analyzer/search.py.
+template <typename F>
+__device__ int wg_binary_search(double x_min, double x_max, double y_target, F eval,
+                                double *x_star) {
+  double y0 = eval(x_min);
+  if (within_tolerance(y0, y_target, kTolerance)) {
+    *x_star = x_min;
+    return 0;
+  }
+  double y1 = eval(x_max);
+  if (within_tolerance(y1, y_target, kTolerance)) {
+    *x_star = x_max;
+    return 0;
+  }
+  bool increasing = y0 < y1;
+  if ((increasing && y_target < y0) || (!increasing && y_target > y0)) {
+    *x_star = x_min;
+    return -1;
+  }
+  if ((increasing && y_target > y1) || (!increasing && y_target < y1)) {
+    *x_star = x_max;
+    return +1;
+  }
+  double xs = x_min;
+  for (int i = 0; i < kMaxIterations; ++i) {
+    xs = 0.5 * (x_min + x_max);
+    double ys = eval(xs);
+    if (within_tolerance(ys, y_target, kTolerance)) break;
+    if ((increasing && y_target < ys) || (!increasing && y_target > ys)) {
+      x_max = xs;
+    } else {
+      x_min = xs;
+    }
+  }
+  *x_star = xs;
+  return 0;
+}
+
+extern "C" __global__ void __launch_bounds__(WVA_THREADS)
+    wva_solve_kernel(const double *__restrict__ prob, double *__restrict__ out,
+                     int n_problems, int max_k) {
+  const int pid = blockIdx.x;
+  if (pid >= n_problems) return;
+  const double *pr = prob + (size_t)pid * PROBLEM_FIELDS;
+  double *res = out + (size_t)pid * RESULT_FIELDS;
+  const int tid = threadIdx.x;
+
+  Parms p;
+  p.alpha = pr[P_ALPHA];
+  p.beta = pr[P_BETA];
+  p.gamma = pr[P_GAMMA];
+  p.delta = pr[P_DELTA];
+  p.in_tokens = pr[P_IN_TOKENS];
+  p.out_tokens = (int)pr[P_OUT_TOKENS];
+  p.max_batch = (int)pr[P_MAX_BATCH];
+  p.num_decode = p.out_tokens - 1;
+  if (p.in_tokens == 0.0 && p.out_tokens == 1) p.num_decode = 1;
+
+  const int K = p.max_batch * (1 + kMaxQueueToBatchRatio);
+
+  extern __shared__ double smem[];
+  double *cum = smem;             // this problem's K entries
+  double *totals = smem + max_k;  // 256 chunk totals
+  double *red = totals + WVA_THREADS;
+
+  // chunked parallel inclusive scan of log_mu over K states
+  const int chunk = (K + WVA_THREADS - 1) / WVA_THREADS;
+  const int lo = tid * chunk;
+  const int hi = min(lo + chunk, K);
+  double acc = 0.0;
+  for (int n = lo; n < hi; ++n) {
+    acc += log_mu(p, n);
+    cum[n] = acc;
+  }
+  totals[tid] = (lo < K) ? acc : 0.0;
+  __syncthreads();
+  double offset = 0.0;
+  for (int t = 0; t < tid; ++t) offset += totals[t];  // broadcast LDS reads
+  for (int n = lo; n < hi; ++n) cum[n] += offset;
+  __syncthreads();
+
+  const double lam_min = serv_rate(p, 1) * kEpsilon;  // req/ms
+  const double lam_max = serv_rate(p, p.max_batch) * (1.0 - kEpsilon);
+
+  WgEval ev{p, cum, red, K};
+
+  if (tid < RESULT_FIELDS && pid < n_problems) res[tid] = 0.0;
+
+  double lam_ttft = lam_max;
+  if (pr[P_TARGET_TTFT] > 0.0) {
+    int ind = wg_binary_search(
+        lam_min, lam_max, pr[P_TARGET_TTFT],
+        [&](double x) { return ev.eval_ttft(x); }, &lam_ttft);
+    if (ind < 0) return;
+  }
+  double lam_itl = lam_max;
+  if (pr[P_TARGET_ITL] > 0.0) {
+    int ind = wg_binary_search(
+        lam_min, lam_max, pr[P_TARGET_ITL],
+        [&](double x) { return ev.eval_itl(x); }, &lam_itl);
+    if (ind < 0) return;
+  }
+  double lam_tps = lam_max;
+  if (pr[P_TARGET_TPS] > 0.0) lam_tps = lam_max * (1.0 - kStabilityFraction);
+
+  const double lam = fmin(lam_ttft, fmin(lam_itl, lam_tps));
+  Stats st = ev.eval(lam);
+  const double rate_star = st.throughput * 1000.0;  // req/s
+
+  const double total_rate = pr[P_TOTAL_RATE];
+  double n_rep = ceil(total_rate / rate_star);
+  if (n_rep < pr[P_MIN_REPLICAS]) n_rep = pr[P_MIN_REPLICAS];
+  const double rate = total_rate / n_rep;
+  if (rate <= 0.0 || rate > lam_max * 1000.0) return;
+
+  Stats fin = ev.eval(rate / 1000.0);
+  double rho = fin.n_serv / (double)p.max_batch;
+  rho = fmin(fmax(rho, 0.0), 1.0);
+
+  if (tid == 0) {
+    res[R_FEASIBLE] = 1.0;
+    res[R_REPLICAS] = n_rep;
+    res[R_RATE_STAR] = rate_star;
+    res[R_ITL] = eval_itl_of(p, fin);
+    res[R_TTFT] = eval_ttft_of(p, fin);
+    res[R_RHO] = rho;
+  }
+}
+
+}  // namespace wva
+
+extern "C" void wva_launch_solve(const double *prob, double *out, int n_problems,
+                                 int max_k, void *stream) {
+  const size_t smem = (size_t)(max_k + WVA_THREADS + 32) * sizeof(double);
+  hipLaunchKernelGGL(wva::wva_solve_kernel, dim3(n_problems), dim3(WVA_THREADS),
+                     smem, (hipStream_t)stream, prob, out, n_problems, max_k);
+}

@@ -1,0 +1,46 @@
+"""Native batched queue-solver ops.
+
+Loads the in-tree extension ``wva_amd._queue_native`` (built by setup.py via
+hipcc for gfx950).  Policy:
+
+- on a machine with a GPU, a missing/broken native extension is a hard
+  error — the HIP path must be the path that actually runs;
+- on CPU-only machines the pure-Python analyzer fallback is allowed (it is
+  the semantic reference the native code is tested against).
+"""
+
+from __future__ import annotations
+
+_native = None
+_native_err: Exception | None = None
+
+try:  # pragma: no cover - import side effect
+    import torch  # noqa: F401  (the extension links against libtorch)
+
+    from wva_amd import _queue_native as _native  # type: ignore
+except Exception as e:  # pragma: no cover
+    _native_err = e
+
+
+def native_available() -> bool:
+    return _native is not None
+
+
+def get_native():
+    """Return the native module, failing loudly when a GPU is present."""
+    if _native is not None:
+        return _native
+    import torch
+
+    if torch.cuda.is_available():
+        raise RuntimeError(
+            "wva_amd._queue_native is not importable on a GPU machine — "
+            "build it with `PYTORCH_ROCM_ARCH=gfx950 python setup.py "
+            f"build_ext --inplace` (import error: {_native_err})"
+        )
+    return None
+
+
+from .batched import BatchedAllocationSolver, solve_problems  # noqa: E402
+
+__all__ = ["native_available", "get_native", "BatchedAllocationSolver", "solve_problems"]

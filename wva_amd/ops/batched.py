@@ -1,0 +1,234 @@
+"""Batched allocation sizing over a whole System.
+
+The reference sizes one (server, accelerator) pair at a time inside the
+reconcile loop (hot loop #1: /root/reference/pkg/core/server.go:55-67 ->
+allocation.go:27-163).  Here the analyze phase is *batched*: all pairs of
+the fleet are packed into one [B, 12] float64 matrix and solved either by
+the native extension (CPU C++/at::parallel_for, or the gfx950 HIP kernel —
+one workgroup per pair) or by the pure-Python analyzer fallback.
+
+Field layouts mirror wva_amd/csrc/queue_core.h.
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Dict, List, Optional, Tuple
+
+import numpy as np
+
+from ..analyzer import (
+    AnalyzerError,
+    Configuration,
+    DecodeParms,
+    PrefillParms,
+    QueueAnalyzer,
+    RequestSize,
+    ServiceParms,
+    TargetPerf,
+)
+from ..config import MAX_QUEUE_TO_BATCH_RATIO
+from ..core import Allocation, System
+from ..core.allocation import _zero_load_allocation
+
+# problem columns
+P_ALPHA, P_BETA, P_GAMMA, P_DELTA = 0, 1, 2, 3
+P_IN_TOKENS, P_OUT_TOKENS, P_MAX_BATCH = 4, 5, 6
+P_TARGET_TTFT, P_TARGET_ITL, P_TARGET_TPS = 7, 8, 9
+P_TOTAL_RATE, P_MIN_REPLICAS = 10, 11
+PROBLEM_FIELDS = 12
+
+# result columns
+R_FEASIBLE, R_REPLICAS, R_RATE_STAR, R_ITL, R_TTFT, R_RHO = range(6)
+RESULT_FIELDS = 6
+
+# the GPU path keeps the cumulative table in LDS: N <= ~700
+GPU_MAX_BATCH_LIMIT = 700
+
+
+def _solve_problems_python(problems: np.ndarray) -> np.ndarray:
+    """Reference-semantics scalar fallback via the Python analyzer."""
+    out = np.zeros((problems.shape[0], RESULT_FIELDS), dtype=np.float64)
+    for i, row in enumerate(problems):
+        N = int(row[P_MAX_BATCH])
+        config = Configuration(
+            max_batch_size=N,
+            max_queue_size=N * MAX_QUEUE_TO_BATCH_RATIO,
+            service_parms=ServiceParms(
+                prefill=PrefillParms(gamma=row[P_GAMMA], delta=row[P_DELTA]),
+                decode=DecodeParms(alpha=row[P_ALPHA], beta=row[P_BETA]),
+            ),
+        )
+        try:
+            qa = QueueAnalyzer(
+                config,
+                RequestSize(
+                    avg_input_tokens=int(row[P_IN_TOKENS]),
+                    avg_output_tokens=int(row[P_OUT_TOKENS]),
+                ),
+            )
+            _, metrics, _ = qa.size(
+                TargetPerf(
+                    target_ttft=row[P_TARGET_TTFT],
+                    target_itl=row[P_TARGET_ITL],
+                    target_tps=row[P_TARGET_TPS],
+                )
+            )
+            rate_star = metrics.throughput
+            total_rate = row[P_TOTAL_RATE]
+            n_rep = max(math.ceil(total_rate / rate_star), int(row[P_MIN_REPLICAS]))
+            final = qa.analyze(total_rate / n_rep)
+        except AnalyzerError:
+            continue
+        out[i, R_FEASIBLE] = 1.0
+        out[i, R_REPLICAS] = n_rep
+        out[i, R_RATE_STAR] = rate_star
+        out[i, R_ITL] = final.avg_token_time
+        out[i, R_TTFT] = final.avg_wait_time + final.avg_prefill_time
+        out[i, R_RHO] = final.rho
+    return out
+
+
+def solve_problems(problems: np.ndarray, device: Optional[str] = None) -> np.ndarray:
+    """Solve a [B, 12] problem matrix -> [B, 6] results.
+
+    device: None for automatic (native CPU if built, else Python), "cpu" to
+    force the native/Python CPU path, or a torch device string like "cuda"
+    to run the gfx950 kernel.
+    """
+    from . import get_native, native_available
+
+    problems = np.ascontiguousarray(problems, dtype=np.float64)
+    if problems.ndim != 2 or problems.shape[1] != PROBLEM_FIELDS:
+        raise ValueError(f"problems must be [B, {PROBLEM_FIELDS}]")
+    if problems.shape[0] == 0:
+        return np.zeros((0, RESULT_FIELDS), dtype=np.float64)
+
+    want_gpu = device is not None and str(device).startswith("cuda")
+    if want_gpu:
+        native = get_native()  # raises loudly if missing on a GPU box
+        import torch
+
+        if problems[:, P_MAX_BATCH].max() > GPU_MAX_BATCH_LIMIT:
+            # LDS-resident limit: solve oversized problems on CPU
+            big = problems[:, P_MAX_BATCH] > GPU_MAX_BATCH_LIMIT
+            out = np.empty((problems.shape[0], RESULT_FIELDS), dtype=np.float64)
+            out[~big] = solve_problems(problems[~big], device=device)
+            out[big] = solve_problems(problems[big], device="cpu")
+            return out
+        t = torch.from_numpy(problems).to(device)
+        res = native.solve_allocations(t)
+        return res.cpu().numpy()
+
+    if native_available():
+        import torch
+
+        native = get_native()
+        return native.solve_allocations(torch.from_numpy(problems)).numpy()
+    return _solve_problems_python(problems)
+
+
+class BatchedAllocationSolver:
+    """Batched analyze phase: computes ``server.all_allocations`` for every
+    server in a System in one native call (semantically equal to
+    ``System.calculate()``)."""
+
+    def __init__(self, device: Optional[str] = None) -> None:
+        self.device = device
+
+    def calculate(self, system: System) -> None:
+        for g in system.accelerators.values():
+            g.calculate()
+
+        rows: List[List[float]] = []
+        keys: List[Tuple[str, str, int]] = []  # (server, acc, N)
+        zero_load: Dict[str, Dict[str, Allocation]] = {}
+        costs: List[float] = []
+
+        for server in system.servers.values():
+            server.all_allocations = {}
+            load = server.load
+            if (
+                load is None
+                or load.arrival_rate < 0
+                or load.avg_in_tokens < 0
+                or load.avg_out_tokens < 0
+            ):
+                continue
+            model = system.model(server.model_name)
+            if model is None:
+                continue
+            svc = system.service_class(server.service_class_name)
+            if svc is None:
+                continue
+            target = svc.model_target(server.model_name)
+            if target is None:
+                continue
+            candidates = server.get_candidate_accelerators(system.accelerators)
+            for acc in candidates.values():
+                perf = model.get_perf_data(acc.name)
+                if perf is None:
+                    continue
+                if load.arrival_rate == 0 or load.avg_out_tokens == 0:
+                    alloc = _zero_load_allocation(server, model, acc, perf)
+                    zero_load.setdefault(server.name, {})[acc.name] = alloc
+                    continue
+                K = int(load.avg_out_tokens)
+                if server.max_batch_size > 0:
+                    N = server.max_batch_size
+                else:
+                    N = max(perf.max_batch_size * perf.at_tokens // K, 1)
+                if target.tps == 0:
+                    total_rate = load.arrival_rate / 60.0
+                else:
+                    total_rate = target.tps / float(K)
+                rows.append(
+                    [
+                        perf.decode_parms.alpha,
+                        perf.decode_parms.beta,
+                        perf.prefill_parms.gamma,
+                        perf.prefill_parms.delta,
+                        float(int(load.avg_in_tokens)),
+                        float(K),
+                        float(N),
+                        target.ttft,
+                        target.itl,
+                        target.tps,
+                        total_rate,
+                        float(server.min_num_replicas),
+                    ]
+                )
+                keys.append((server.name, acc.name, N))
+                costs.append(acc.cost * model.get_num_instances(acc.name))
+
+        if rows:
+            results = solve_problems(np.asarray(rows, dtype=np.float64), self.device)
+        else:
+            results = np.zeros((0, RESULT_FIELDS))
+
+        for (server_name, acc_name, N), res, cost_per_replica in zip(keys, results, costs):
+            if res[R_FEASIBLE] != 1.0:
+                continue
+            server = system.server(server_name)
+            num_replicas = int(res[R_REPLICAS])
+            alloc = Allocation(
+                accelerator=acc_name,
+                num_replicas=num_replicas,
+                batch_size=N,
+                cost=cost_per_replica * num_replicas,
+                itl=float(res[R_ITL]),
+                ttft=float(res[R_TTFT]),
+                rho=float(res[R_RHO]),
+                max_arrv_rate_per_replica=float(res[R_RATE_STAR]) / 1000.0,
+            )
+            alloc.set_value(alloc.cost)
+            if server.cur_allocation is not None:
+                alloc.set_value(server.cur_allocation.transition_penalty(alloc))
+            server.all_allocations[acc_name] = alloc
+
+        for server_name, accs in zero_load.items():
+            server = system.server(server_name)
+            for acc_name, alloc in accs.items():
+                if server.cur_allocation is not None:
+                    alloc.set_value(server.cur_allocation.transition_penalty(alloc))
+                server.all_allocations[acc_name] = alloc
