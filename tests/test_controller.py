@@ -1,0 +1,320 @@
+"""Component tests for the controller stack with a fake API server and
+MockPromAPI — the envtest-tier analog (controller,
+collector, optimizer, actuator suites)."""
+
+import pytest
+from prometheus_client import CollectorRegistry
+
+from wva_amd.api import v1alpha1
+from wva_amd.controller import collector, metrics as ctrl_metrics
+from wva_amd.controller.promclient import MockPromAPI, PromQueryError
+from wva_amd.controller.reconciler import (
+    CONFIG_MAP_NAMESPACE,
+    SERVICE_CLASSES_CM,
+    ManagerRuntime,
+    VariantAutoscalingReconciler,
+    parse_go_duration,
+)
+from wva_amd.controller.utils import find_model_slo
+from wva_amd.kube import ConfigMap, Deployment, NotFoundError
+from kube_fixtures import (
+    make_cluster,
+    make_deployment,
+    make_va,
+    set_load_metrics,
+)
+
+
+@pytest.fixture()
+def registry():
+    reg = CollectorRegistry()
+    ctrl_metrics.init_metrics(reg)
+    yield reg
+    ctrl_metrics.reset_metrics()
+
+
+@pytest.fixture()
+def cluster():
+    return make_cluster()
+
+
+@pytest.fixture()
+def prom():
+    return MockPromAPI()
+
+
+def get_va(client, name="vllm-llama", namespace="default"):
+    return client.get(v1alpha1.VariantAutoscaling, name, namespace)
+
+
+class TestDurationParsing:
+    def test_basic(self):
+        assert parse_go_duration("60s") == 60.0
+        assert parse_go_duration("1m30s") == 90.0
+        assert parse_go_duration("100ms") == 0.1
+        assert parse_go_duration("2h") == 7200.0
+
+    def test_invalid(self):
+        for bad in ("", "abc", "10", "5x"):
+            with pytest.raises(ValueError):
+                parse_go_duration(bad)
+
+
+class TestReconcileHappyPath:
+    def test_full_cycle(self, cluster, prom, registry):
+        make_deployment(cluster, replicas=2)
+        make_va(cluster)
+        set_load_metrics(prom, "default/llama-8b", "default", arrival_rps=2.0)
+
+        rec = VariantAutoscalingReconciler(cluster, prom)
+        result = rec.reconcile()
+        assert result.requeue_after == 1.0  # GLOBAL_OPT_INTERVAL=1s
+
+        va = get_va(cluster)
+        # current allocation scraped from Prometheus
+        assert va.status.current_alloc.num_replicas == 2
+        assert va.status.current_alloc.accelerator == "MI355X"
+        assert float(va.status.current_alloc.load.arrival_rate) == pytest.approx(120.0)  # req/min
+        assert va.status.current_alloc.max_batch == 256
+        assert float(va.status.current_alloc.variant_cost) == pytest.approx(170.0)
+        # optimization produced a desired allocation
+        assert va.status.desired_optimized_alloc.accelerator == "MI355X"
+        assert va.status.desired_optimized_alloc.num_replicas >= 1
+        # conditions
+        assert v1alpha1.is_condition_true(va, v1alpha1.TYPE_METRICS_AVAILABLE)
+        assert v1alpha1.is_condition_true(va, v1alpha1.TYPE_OPTIMIZATION_READY)
+        assert va.status.actuation.applied
+
+        # ownerReference set for GC
+        assert any(r.kind == "Deployment" and r.controller for r in va.metadata.owner_references)
+
+        # inferno_* gauges emitted for HPA/KEDA
+        labels = {"variant_name": "vllm-llama", "namespace": "default", "accelerator_type": "MI355X"}
+        assert registry.get_sample_value("inferno_current_replicas", labels) == 2.0
+        desired = registry.get_sample_value("inferno_desired_replicas", labels)
+        assert desired == float(va.status.desired_optimized_alloc.num_replicas)
+        assert registry.get_sample_value("inferno_desired_ratio", labels) == desired / 2.0
+        # the solver latency histogram observed this cycle
+        assert registry.get_sample_value("wva_solver_duration_seconds_count") == 1.0
+
+    def test_high_load_scales_out(self, cluster, prom, registry):
+        make_deployment(cluster, replicas=1)
+        make_va(cluster)
+        # 20 req/s of 200-token outputs (mirrors the reference optimizer
+        # suite's injected high load, optimizer_test.go:337-459)
+        set_load_metrics(
+            prom, "default/llama-8b", "default", arrival_rps=20.0, out_tokens=200.0
+        )
+        VariantAutoscalingReconciler(cluster, prom).reconcile()
+        va = get_va(cluster)
+        assert va.status.desired_optimized_alloc.num_replicas > 1
+
+    def test_zero_load_min_replicas(self, cluster, prom, registry):
+        make_deployment(cluster, replicas=1)
+        make_va(cluster)
+        set_load_metrics(prom, "default/llama-8b", "default", arrival_rps=0.0, out_tokens=0.0)
+        VariantAutoscalingReconciler(cluster, prom).reconcile()
+        va = get_va(cluster)
+        assert va.status.desired_optimized_alloc.num_replicas == 1
+
+    def test_scale_to_zero_env(self, cluster, prom, registry, monkeypatch):
+        monkeypatch.setenv("WVA_SCALE_TO_ZERO", "true")
+        make_deployment(cluster, replicas=1)
+        make_va(cluster)
+        set_load_metrics(prom, "default/llama-8b", "default", arrival_rps=0.0, out_tokens=0.0)
+        VariantAutoscalingReconciler(cluster, prom).reconcile()
+        va = get_va(cluster)
+        assert va.status.desired_optimized_alloc.num_replicas == 0
+
+
+class TestGracefulDegradation:
+    def test_missing_configmap_fails_cycle(self, prom, registry):
+        from wva_amd.kube import InMemoryKubeClient
+
+        client = InMemoryKubeClient()
+        rec = VariantAutoscalingReconciler(client, prom)
+        with pytest.raises(NotFoundError):
+            rec.reconcile()
+
+    def test_metrics_missing_skips_variant(self, cluster, prom, registry):
+        make_deployment(cluster)
+        make_va(cluster)
+        # availability probe returns empty for both query forms
+        model = "default/llama-8b"
+        prom.query_results[
+            f'vllm:request_success_total{{model_name="{model}",namespace="default"}}'
+        ] = []
+        prom.query_results[f'vllm:request_success_total{{model_name="{model}"}}'] = []
+        VariantAutoscalingReconciler(cluster, prom).reconcile()
+        va = get_va(cluster)
+        # VA untouched: no optimized alloc, no conditions persisted
+        assert va.status.desired_optimized_alloc.num_replicas == 0
+        assert va.status.desired_optimized_alloc.accelerator == ""
+
+    def test_stale_metrics_skip(self, cluster, prom, registry):
+        make_deployment(cluster)
+        make_va(cluster)
+        model = "default/llama-8b"
+        prom.set_result(
+            f'vllm:request_success_total{{model_name="{model}",namespace="default"}}',
+            5.0,
+            age_seconds=600.0,
+        )
+        VariantAutoscalingReconciler(cluster, prom).reconcile()
+        va = get_va(cluster)
+        assert va.status.desired_optimized_alloc.accelerator == ""
+
+    def test_prometheus_error_skips_variant(self, cluster, prom, registry):
+        make_deployment(cluster)
+        make_va(cluster)
+        model = "default/llama-8b"
+        prom.set_error(
+            f'vllm:request_success_total{{model_name="{model}",namespace="default"}}',
+            RuntimeError("boom"),
+        )
+        VariantAutoscalingReconciler(cluster, prom).reconcile()
+        va = get_va(cluster)
+        assert va.status.desired_optimized_alloc.accelerator == ""
+
+    def test_one_bad_variant_does_not_block_others(self, cluster, prom, registry):
+        make_deployment(cluster, name="good")
+        make_va(cluster, name="good")
+        make_va(cluster, name="orphan")  # no Deployment
+        set_load_metrics(prom, "default/llama-8b", "default")
+        VariantAutoscalingReconciler(cluster, prom).reconcile()
+        assert get_va(cluster, "good").status.desired_optimized_alloc.num_replicas >= 1
+        assert get_va(cluster, "orphan").status.desired_optimized_alloc.accelerator == ""
+
+    def test_deleted_variant_filtered(self, cluster, prom, registry):
+        import datetime
+
+        make_deployment(cluster)
+        va = make_va(cluster)
+        va.metadata.deletion_timestamp = datetime.datetime.now(datetime.timezone.utc)
+        cluster.update(va)
+        result = VariantAutoscalingReconciler(cluster, prom).reconcile()
+        # nothing active: no requeue change, no optimization
+        assert result.requeue_after is None
+
+    def test_missing_accelerator_cost_skips(self, cluster, prom, registry):
+        make_deployment(cluster)
+        make_va(cluster, accelerator="H100")  # not in the unit-cost table
+        set_load_metrics(prom, "default/llama-8b", "default")
+        VariantAutoscalingReconciler(cluster, prom).reconcile()
+        va = get_va(cluster)
+        assert va.status.desired_optimized_alloc.accelerator == ""
+
+
+class TestOwnerRefGC:
+    def test_va_garbage_collected_on_deployment_delete(self, cluster, prom, registry):
+        make_deployment(cluster)
+        make_va(cluster)
+        set_load_metrics(prom, "default/llama-8b", "default")
+        VariantAutoscalingReconciler(cluster, prom).reconcile()
+        assert get_va(cluster) is not None
+        cluster.delete(Deployment, "vllm-llama", "default")
+        with pytest.raises(NotFoundError):
+            get_va(cluster)
+
+
+class TestFindModelSLO:
+    def test_found(self, cluster):
+        cm = cluster.get(ConfigMap, SERVICE_CLASSES_CM, CONFIG_MAP_NAMESPACE)
+        entry, cls = find_model_slo(cm.data, "default/llama-8b")
+        assert cls == "Premium"
+        assert entry.slo_tpot == 24
+        assert entry.slo_ttft == 500
+
+    def test_not_found(self, cluster):
+        cm = cluster.get(ConfigMap, SERVICE_CLASSES_CM, CONFIG_MAP_NAMESPACE)
+        with pytest.raises(KeyError):
+            find_model_slo(cm.data, "nope")
+
+
+class TestManagerRuntime:
+    def test_env_precedence_over_configmap(self, cluster, prom, monkeypatch):
+        monkeypatch.setenv("PROMETHEUS_BASE_URL", "https://env-prom:9090")
+        runtime = ManagerRuntime.__new__(ManagerRuntime)
+        runtime.client = cluster
+        config = runtime._get_prometheus_config()
+        assert config.base_url == "https://env-prom:9090"
+
+    def test_configmap_fallback(self, cluster, prom, monkeypatch):
+        monkeypatch.delenv("PROMETHEUS_BASE_URL", raising=False)
+        runtime = ManagerRuntime.__new__(ManagerRuntime)
+        runtime.client = cluster
+        config = runtime._get_prometheus_config()
+        assert config.base_url == "https://prom.test:9090"
+
+    def test_https_mandatory(self, cluster, prom):
+        from wva_amd.controller.interfaces import PrometheusConfig
+        from wva_amd.controller.promclient import validate_tls_config
+
+        with pytest.raises(ValueError):
+            validate_tls_config(PrometheusConfig(base_url="http://insecure:9090"))
+
+    def test_create_event_wakes_loop(self, cluster, prom, registry):
+        runtime = ManagerRuntime(cluster, prom_api=prom)
+        assert not runtime._wake.is_set()
+        make_va(cluster, name="fresh")
+        assert runtime._wake.is_set()
+        runtime._wake.clear()
+        # non-watched configmap does not wake
+        from wva_amd.api.v1alpha1.types import ObjectMeta
+
+        cluster.create(ConfigMap(metadata=ObjectMeta(name="other", namespace="x")))
+        assert not runtime._wake.is_set()
+
+    def test_run_respects_max_cycles(self, cluster, prom, registry):
+        make_deployment(cluster)
+        make_va(cluster)
+        set_load_metrics(prom, "default/llama-8b", "default")
+        runtime = ManagerRuntime(cluster, prom_api=prom)
+        runtime.run(max_cycles=2)
+        assert registry.get_sample_value("wva_solver_duration_seconds_count") == 2.0
+
+
+class TestCollectorQueries:
+    def test_query_shapes(self):
+        q = collector.arrival_query("m", "ns")
+        assert q == 'sum(rate(vllm:request_success_total{model_name="m",namespace="ns"}[1m]))'
+        q = collector.itl_query("m", "ns")
+        assert (
+            q
+            == 'sum(rate(vllm:time_per_output_token_seconds_sum{model_name="m",namespace="ns"}[1m]))'
+            '/sum(rate(vllm:time_per_output_token_seconds_count{model_name="m",namespace="ns"}[1m]))'
+        )
+
+    def test_nan_inf_fixed(self, prom):
+        from wva_amd.controller.collector import _query_value
+
+        prom.set_result("q1", float("nan"))
+        prom.set_result("q2", float("inf"))
+        assert _query_value(prom, "q1", "x") == 0.0
+        assert _query_value(prom, "q2", "x") == 0.0
+
+    def test_emulator_fallback_validation(self, prom):
+        model, ns = "emu-model", "default"
+        prom.query_results[
+            f'vllm:request_success_total{{model_name="{model}",namespace="{ns}"}}'
+        ] = []
+        # fallback (no namespace) left at the default fresh zero sample
+        result = collector.validate_metrics_availability(prom, model, ns)
+        assert result.available
+        assert result.reason == v1alpha1.REASON_METRICS_FOUND
+
+
+class TestBatchedReconcile:
+    def test_batched_analyzer_matches_scalar(self, cluster, prom, registry):
+        make_deployment(cluster, replicas=1)
+        make_va(cluster)
+        set_load_metrics(prom, "default/llama-8b", "default", arrival_rps=20.0, out_tokens=200.0)
+        VariantAutoscalingReconciler(cluster, prom).reconcile()
+        scalar_desired = get_va(cluster).status.desired_optimized_alloc.num_replicas
+
+        cluster2 = make_cluster()
+        make_deployment(cluster2, replicas=1)
+        make_va(cluster2)
+        VariantAutoscalingReconciler(cluster2, prom, batched_analyzer=True).reconcile()
+        batched_desired = get_va(cluster2).status.desired_optimized_alloc.num_replicas
+        assert batched_desired == scalar_desired > 1

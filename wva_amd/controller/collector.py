@@ -1,0 +1,228 @@
+"""Prometheus collector: vLLM serving metrics + MI355X GPU telemetry.
+
+Parity with /root/reference/internal/collector/collector.go:
+- availability validation with emulator fallback (namespace-less query) and
+  a 5-minute staleness gate (collector.go:87-156);
+- the five 1-minute-rate PromQL shapes (collector.go:170-209) with the
+  req/s->req/min and s->ms unit conversions (collector.go:217,233,239);
+- NaN/Inf values fixed to 0; maxBatch hardcoded 256 pending server-reported
+  values (collector.go:258-259).
+
+MI355X addition: ``collect_gpu_telemetry`` scrapes amd-smi/rocm-smi
+exporter series (utilization, VRAM, power) as auxiliary signals — there is
+no NVML/DCGM code path anywhere.
+"""
+
+from __future__ import annotations
+
+import math
+import time
+from dataclasses import dataclass
+from typing import Dict, Optional
+
+from ..api import v1alpha1
+from ..kube import Deployment
+from . import constants
+from .logger import log
+from .promclient import PromAPI, PromQueryError, Sample
+
+STALENESS_LIMIT_SECONDS = 5 * 60
+
+# vendor resource prefixes for (future) limited-mode inventory; AMD first
+VENDORS = ["amd.com", "nvidia.com", "intel.com"]
+
+
+@dataclass
+class MetricsValidationResult:
+    available: bool
+    reason: str
+    message: str
+
+
+def _fix_value(x: float) -> float:
+    return 0.0 if (math.isnan(x) or math.isinf(x)) else x
+
+
+def _query_value(prom: PromAPI, query: str, metric_name: str) -> float:
+    try:
+        vec = prom.query(query)
+    except PromQueryError as e:
+        raise PromQueryError(f"failed to query Prometheus for {metric_name}: {e}") from e
+    if not vec:
+        return 0.0
+    return _fix_value(vec[0].value)
+
+
+# ------------------------------------------------------------ query builders
+def arrival_query(model: str, namespace: str) -> str:
+    return (
+        f'sum(rate({constants.VLLM_REQUEST_SUCCESS_TOTAL}'
+        f'{{{constants.LABEL_MODEL_NAME}="{model}",{constants.LABEL_NAMESPACE}="{namespace}"}}[1m]))'
+    )
+
+
+def _ratio_query(sum_metric: str, count_metric: str, model: str, namespace: str) -> str:
+    sel = f'{{{constants.LABEL_MODEL_NAME}="{model}",{constants.LABEL_NAMESPACE}="{namespace}"}}'
+    return f"sum(rate({sum_metric}{sel}[1m]))/sum(rate({count_metric}{sel}[1m]))"
+
+
+def avg_prompt_tokens_query(model: str, namespace: str) -> str:
+    return _ratio_query(
+        constants.VLLM_REQUEST_PROMPT_TOKENS_SUM,
+        constants.VLLM_REQUEST_PROMPT_TOKENS_COUNT,
+        model,
+        namespace,
+    )
+
+
+def avg_generation_tokens_query(model: str, namespace: str) -> str:
+    return _ratio_query(
+        constants.VLLM_REQUEST_GENERATION_TOKENS_SUM,
+        constants.VLLM_REQUEST_GENERATION_TOKENS_COUNT,
+        model,
+        namespace,
+    )
+
+
+def ttft_query(model: str, namespace: str) -> str:
+    return _ratio_query(
+        constants.VLLM_TIME_TO_FIRST_TOKEN_SECONDS_SUM,
+        constants.VLLM_TIME_TO_FIRST_TOKEN_SECONDS_COUNT,
+        model,
+        namespace,
+    )
+
+
+def itl_query(model: str, namespace: str) -> str:
+    return _ratio_query(
+        constants.VLLM_TIME_PER_OUTPUT_TOKEN_SECONDS_SUM,
+        constants.VLLM_TIME_PER_OUTPUT_TOKEN_SECONDS_COUNT,
+        model,
+        namespace,
+    )
+
+
+# ------------------------------------------------------------- availability
+def validate_metrics_availability(
+    prom: PromAPI, model_name: str, namespace: str
+) -> MetricsValidationResult:
+    """Check that vLLM metrics exist and are fresh for (model, namespace)."""
+    test_query = (
+        f'{constants.VLLM_REQUEST_SUCCESS_TOTAL}'
+        f'{{{constants.LABEL_MODEL_NAME}="{model_name}",{constants.LABEL_NAMESPACE}="{namespace}"}}'
+    )
+    try:
+        vec = prom.query(test_query)
+    except PromQueryError as e:
+        log.error("Error querying Prometheus for metrics validation", model=model_name, error=str(e))
+        return MetricsValidationResult(
+            False, v1alpha1.REASON_PROMETHEUS_ERROR, f"Failed to query Prometheus: {e}"
+        )
+
+    if not vec:
+        # emulator fallback: no namespace label
+        fallback = (
+            f'{constants.VLLM_REQUEST_SUCCESS_TOTAL}'
+            f'{{{constants.LABEL_MODEL_NAME}="{model_name}"}}'
+        )
+        try:
+            vec = prom.query(fallback)
+        except PromQueryError as e:
+            return MetricsValidationResult(
+                False, v1alpha1.REASON_PROMETHEUS_ERROR, f"Failed to query Prometheus: {e}"
+            )
+        if not vec:
+            return MetricsValidationResult(
+                False,
+                v1alpha1.REASON_METRICS_MISSING,
+                f"No vLLM metrics found for model '{model_name}' in namespace "
+                f"'{namespace}'. Check: (1) ServiceMonitor exists, (2) selector "
+                f"matches the vLLM service, (3) vLLM pods expose /metrics, "
+                f"(4) Prometheus scrapes the monitoring namespace",
+            )
+
+    now = time.time()
+    for sample in vec:
+        ts = sample.timestamp or now
+        age = now - ts
+        if age > STALENESS_LIMIT_SECONDS:
+            return MetricsValidationResult(
+                False,
+                v1alpha1.REASON_METRICS_STALE,
+                f"vLLM metrics for model '{model_name}' are stale "
+                f"(last update: {age:.0f}s ago)",
+            )
+    return MetricsValidationResult(
+        True, v1alpha1.REASON_METRICS_FOUND, "vLLM metrics are available and up-to-date"
+    )
+
+
+# ----------------------------------------------------------------- main path
+ACCELERATOR_LABEL = "inference.optimization/acceleratorName"
+
+# TODO(parity): collect the live max batch size from the server
+# (collector.go:258-259 hardcodes 256 with the same TODO)
+DEFAULT_MAX_BATCH = 256
+
+
+def add_metrics_to_opt_status(
+    va: v1alpha1.VariantAutoscaling,
+    deployment: Deployment,
+    accelerator_cost: float,
+    prom: PromAPI,
+) -> v1alpha1.Allocation:
+    """Scrape the five vLLM signals and build status.currentAlloc."""
+    namespace = deployment.namespace
+    model = va.spec.model_id
+
+    arrival = _query_value(prom, arrival_query(model, namespace), "ArrivalRate") * 60.0
+    avg_in = _query_value(prom, avg_prompt_tokens_query(model, namespace), "AvgInputTokens")
+    avg_out = _query_value(prom, avg_generation_tokens_query(model, namespace), "AvgOutputTokens")
+    ttft_ms = _query_value(prom, ttft_query(model, namespace), "TTFTAverageTime") * 1000.0
+    itl_ms = _query_value(prom, itl_query(model, namespace), "ITLAverage") * 1000.0
+
+    num_replicas = int(deployment.spec.replicas or 0)
+    acc = va.metadata.labels.get(ACCELERATOR_LABEL, "")
+    if not acc:
+        log.warn("acceleratorName label not found on VariantAutoscaling object", name=va.name)
+    cost = num_replicas * accelerator_cost
+
+    return v1alpha1.Allocation(
+        accelerator=acc,
+        numReplicas=num_replicas,
+        maxBatch=DEFAULT_MAX_BATCH,
+        variantCost=f"{cost:.2f}",
+        ttftAverage=f"{ttft_ms:.2f}",
+        itlAverage=f"{itl_ms:.2f}",
+        load=v1alpha1.LoadProfile(
+            arrivalRate=f"{arrival:.2f}",
+            avgInputTokens=f"{avg_in:.2f}",
+            avgOutputTokens=f"{avg_out:.2f}",
+        ),
+    )
+
+
+# ----------------------------------------------------- MI355X GPU telemetry
+@dataclass
+class GpuTelemetry:
+    utilization_pct: float = 0.0
+    vram_used_bytes: float = 0.0
+    power_watts: float = 0.0
+
+
+def collect_gpu_telemetry(prom: PromAPI, namespace: str) -> Optional[GpuTelemetry]:
+    """Auxiliary amd-smi exporter signals (best-effort; None when absent)."""
+    sel = f'{{{constants.LABEL_NAMESPACE}="{namespace}"}}'
+    try:
+        util = _query_value(prom, f"avg({constants.AMD_SMI_GPU_UTILIZATION}{sel})", "GpuUtil")
+        vram = _query_value(prom, f"sum({constants.AMD_SMI_GPU_VRAM_USED}{sel})", "GpuVram")
+        power = _query_value(prom, f"sum({constants.AMD_SMI_GPU_POWER}{sel})", "GpuPower")
+    except PromQueryError:
+        return None
+    return GpuTelemetry(utilization_pct=util, vram_used_bytes=vram, power_watts=power)
+
+
+def collect_inventory_k8s(client) -> Dict[str, Dict[str, object]]:
+    """Limited-mode inventory stub (collector.go:37-42 keeps the same stub);
+    unlimited mode needs no cluster inventory."""
+    return {}

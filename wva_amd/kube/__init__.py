@@ -1,0 +1,31 @@
+"""Minimal Kubernetes client abstraction.
+
+The reference is a kubebuilder controller talking to a real API server via
+controller-runtime.  This rebuild targets the same protocol surface (typed
+get/list/patch/status-update with NotFound/Invalid/Forbidden error classes,
+ownerReference garbage collection, resourceVersion bumping) behind a small
+``KubeClient`` protocol with two implementations:
+
+- :class:`InMemoryKubeClient` — the envtest analog used by the component
+  and e2e test tiers;
+- a real HTTP client can be slotted in later without touching the
+  controller (the controller only sees the protocol).
+"""
+
+from .errors import ConflictError, ForbiddenError, InvalidError, KubeError, NotFoundError
+from .objects import ConfigMap, Deployment, DeploymentSpec, DeploymentStatus
+from .client import InMemoryKubeClient, KubeClient
+
+__all__ = [
+    "KubeError",
+    "NotFoundError",
+    "InvalidError",
+    "ForbiddenError",
+    "ConflictError",
+    "ConfigMap",
+    "Deployment",
+    "DeploymentSpec",
+    "DeploymentStatus",
+    "KubeClient",
+    "InMemoryKubeClient",
+]
