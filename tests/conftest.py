@@ -1,8 +1,11 @@
 import sys
 from pathlib import Path
 
-# allow `import wva_amd` from a source checkout without installation
-sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
+# allow `import wva_amd` (repo root) and `import vllm_emulator` (tools/)
+# from a source checkout without installation
+_ROOT = Path(__file__).resolve().parent.parent
+sys.path.insert(0, str(_ROOT))
+sys.path.insert(0, str(_ROOT / "tools"))
 
 
 def pytest_configure(config):

@@ -32,6 +32,15 @@ STALENESS_LIMIT_SECONDS = 5 * 60
 VENDORS = ["amd.com", "nvidia.com", "intel.com"]
 
 
+def rate_window() -> str:
+    """PromQL rate window for the load queries.  The reference hardcodes
+    [1m] (collector.go:170-209); here it is configurable via
+    WVA_RATE_WINDOW for fast-cadence deployments and tests."""
+    import os
+
+    return os.environ.get("WVA_RATE_WINDOW", "1m")
+
+
 @dataclass
 class MetricsValidationResult:
     available: bool
@@ -57,13 +66,14 @@ def _query_value(prom: PromAPI, query: str, metric_name: str) -> float:
 def arrival_query(model: str, namespace: str) -> str:
     return (
         f'sum(rate({constants.VLLM_REQUEST_SUCCESS_TOTAL}'
-        f'{{{constants.LABEL_MODEL_NAME}="{model}",{constants.LABEL_NAMESPACE}="{namespace}"}}[1m]))'
+        f'{{{constants.LABEL_MODEL_NAME}="{model}",{constants.LABEL_NAMESPACE}="{namespace}"}}[{rate_window()}]))'
     )
 
 
 def _ratio_query(sum_metric: str, count_metric: str, model: str, namespace: str) -> str:
     sel = f'{{{constants.LABEL_MODEL_NAME}="{model}",{constants.LABEL_NAMESPACE}="{namespace}"}}'
-    return f"sum(rate({sum_metric}{sel}[1m]))/sum(rate({count_metric}{sel}[1m]))"
+    w = rate_window()
+    return f"sum(rate({sum_metric}{sel}[{w}]))/sum(rate({count_metric}{sel}[{w}]))"
 
 
 def avg_prompt_tokens_query(model: str, namespace: str) -> str:

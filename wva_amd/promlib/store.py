@@ -1,0 +1,63 @@
+"""In-memory time-series store with bounded retention."""
+
+from __future__ import annotations
+
+import collections
+import threading
+import time
+from typing import Deque, Dict, FrozenSet, Iterable, List, Tuple
+
+LabelSet = FrozenSet[Tuple[str, str]]
+
+DEFAULT_RETENTION_SECONDS = 15 * 60
+
+
+class Series:
+    __slots__ = ("name", "labels", "samples")
+
+    def __init__(self, name: str, labels: LabelSet) -> None:
+        self.name = name
+        self.labels = labels
+        self.samples: Deque[Tuple[float, float]] = collections.deque()
+
+    def latest_in(self, start: float, end: float):
+        for t, v in reversed(self.samples):
+            if t <= end:
+                return (t, v) if t >= start else None
+        return None
+
+    def range(self, start: float, end: float) -> List[Tuple[float, float]]:
+        return [(t, v) for t, v in self.samples if start <= t <= end]
+
+
+class TimeSeriesStore:
+    def __init__(self, retention: float = DEFAULT_RETENTION_SECONDS) -> None:
+        self._series: Dict[Tuple[str, LabelSet], Series] = {}
+        self._retention = retention
+        self._lock = threading.RLock()
+
+    def add_sample(self, name: str, labels: Dict[str, str], value: float, ts: float = None) -> None:
+        ts = time.time() if ts is None else ts
+        key = (name, frozenset(labels.items()))
+        with self._lock:
+            series = self._series.get(key)
+            if series is None:
+                series = Series(name, key[1])
+                self._series[key] = series
+            series.samples.append((ts, value))
+            cutoff = ts - self._retention
+            while series.samples and series.samples[0][0] < cutoff:
+                series.samples.popleft()
+
+    def select(self, name: str, matchers: Dict[str, str]) -> List[Series]:
+        want = set(matchers.items())
+        with self._lock:
+            return [
+                s
+                for (n, _), s in self._series.items()
+                if n == name and want <= set(s.labels)
+            ]
+
+    def names(self) -> Iterable[str]:
+        with self._lock:
+            return sorted({n for (n, _) in self._series})
