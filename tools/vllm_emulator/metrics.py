@@ -56,6 +56,17 @@ class EmulatorMetrics:
             "vllm:request_generation_tokens", "Generated token count per request.",
             labels, registry=reg, buckets=TOKEN_BUCKETS)
 
+        # pre-register the labeled children so every series is exported
+        # from process start (as real vLLM does) — otherwise the collector's
+        # availability probe sees no series until the first request
+        for metric in (
+            self.running, self.waiting, self.kv_cache_usage, self.request_arrival,
+            self.request_success, self.tokens_total, self.time_per_output_token,
+            self.time_to_first_token, self.queue_time, self.prompt_tokens,
+            self.generation_tokens,
+        ):
+            metric.labels(model_name=model_name)
+
     def expose(self) -> bytes:
         return generate_latest(self.registry)
 
