@@ -1,0 +1,286 @@
+"""Flagship benchmark: global autoscaling solve on a synthetic MI355X fleet.
+
+Measures the BASELINE.json headline metric — solver wall-clock (ms) with
+SLO-attainment (%) — for N models x M accelerator types (config #3: 3
+models x 2 service classes on MI355X nodes, heterogeneous pool of 3
+accelerator types).  One *step* is one full optimization cycle exactly as
+the reconciler runs it: build the System from the cycle's observed loads,
+size every (server, accelerator) pair (the analytic hot loop, batched on
+the GPU via the gfx950 queue-solver kernel when available), run the
+global min-cost solve, aggregate by type and export the solution.
+
+Scaling is WEAK: each rank owns a fixed shard of VARIANTS_PER_GPU variants
+(a fleet shard), so N GPUs optimize an N-times-larger fleet; the reported
+per-step wall-clock is the max over ranks after a barrier, i.e. the time
+for the WHOLE fleet's global solve.
+
+Usage: python bench.py [--gpus N] [--steps K] [--warmup W]
+(launched by torchrun for N>1; reads RANK/WORLD_SIZE/MASTER_* from env).
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+from pathlib import Path
+
+sys.path.insert(0, str(Path(__file__).resolve().parent))
+
+import numpy as np
+import torch
+
+from wva_amd.config import (
+    AcceleratorData,
+    AllocationData,
+    CapacityData,
+    ModelData,
+    OptimizerData,
+    OptimizerSpec,
+    ServerData,
+    ServerLoadSpec,
+    ServerSpec,
+    ServiceClassData,
+    ServiceClassSpec,
+    ModelTarget,
+    SystemSpec,
+)
+from wva_amd.config.mi355x import MI355X_CATALOG
+from wva_amd.config.types import DecodeParmsSpec, ModelAcceleratorPerfData, PrefillParmsSpec
+from wva_amd.core import System
+from wva_amd.solver import Manager, Optimizer
+
+VARIANTS_PER_GPU = 64  # fleet shard per rank (weak scaling)
+MODELS = ["llama-8b", "llama-70b", "mixtral-8x7b"]
+CLASSES = [("Premium", 1), ("Freemium", 10)]
+ACCELERATORS = list(MI355X_CATALOG)  # MI355X, MI300X, L40S
+
+# per-(model, accelerator) linear perf profiles (random-init per BASELINE:
+# deterministic seeded values of realistic magnitude; no network for real
+# fitted profiles)
+_PROFILE_RNG = np.random.default_rng(42)
+PROFILES = {}
+for model_name in MODELS:
+    scale = {"llama-8b": 1.0, "llama-70b": 3.0, "mixtral-8x7b": 2.0}[model_name]
+    for acc_i, acc in enumerate(ACCELERATORS):
+        slow = {"MI355X": 1.0, "MI300X": 1.6, "L40S": 6.0}[acc]
+        PROFILES[(model_name, acc)] = dict(
+            alpha=float(3.0 * scale * slow * _PROFILE_RNG.uniform(0.9, 1.1)),
+            beta=float(0.03 * scale * slow * _PROFILE_RNG.uniform(0.9, 1.1)),
+            gamma=float(8.0 * scale * slow * _PROFILE_RNG.uniform(0.9, 1.1)),
+            delta=float(0.0004 * scale * slow * _PROFILE_RNG.uniform(0.9, 1.1)),
+            acc_count={"llama-8b": 1, "llama-70b": 4, "mixtral-8x7b": 2}[model_name],
+        )
+
+SLOS = {
+    ("Premium", "llama-8b"): (20.0, 800.0),
+    ("Premium", "llama-70b"): (40.0, 2000.0),
+    ("Premium", "mixtral-8x7b"): (30.0, 1500.0),
+    ("Freemium", "llama-8b"): (60.0, 4000.0),
+    ("Freemium", "llama-70b"): (120.0, 8000.0),
+    ("Freemium", "mixtral-8x7b"): (90.0, 6000.0),
+}
+
+
+def build_system_spec(rank: int, step: int, n_variants: int) -> SystemSpec:
+    """Synthetic cluster state for this rank at this step (loads vary per
+    step so every cycle is a fresh solve)."""
+    rng = np.random.default_rng(10_000 * (rank + 1) + step)
+    servers = []
+    for i in range(n_variants):
+        model = MODELS[i % len(MODELS)]
+        cls = CLASSES[(i // len(MODELS)) % len(CLASSES)][0]
+        servers.append(
+            ServerSpec(
+                name=f"va-{rank}-{i}:bench",
+                class_name=cls,
+                model=model,
+                keep_accelerator=False,
+                min_num_replicas=1,
+                max_batch_size=256,
+                current_alloc=AllocationData(
+                    accelerator="MI355X",
+                    num_replicas=int(rng.integers(1, 8)),
+                    load=ServerLoadSpec(
+                        arrival_rate=float(rng.uniform(30.0, 36000.0)),  # req/min
+                        avg_in_tokens=int(rng.integers(64, 2048)),
+                        avg_out_tokens=int(rng.integers(32, 1024)),
+                    ),
+                ),
+            )
+        )
+    return SystemSpec(
+        accelerators=AcceleratorData(spec=list(MI355X_CATALOG.values())),
+        models=ModelData(
+            perf_data=[
+                ModelAcceleratorPerfData(
+                    name=m,
+                    acc=a,
+                    acc_count=PROFILES[(m, a)]["acc_count"],
+                    max_batch_size=256,
+                    decode_parms=DecodeParmsSpec(
+                        alpha=PROFILES[(m, a)]["alpha"], beta=PROFILES[(m, a)]["beta"]
+                    ),
+                    prefill_parms=PrefillParmsSpec(
+                        gamma=PROFILES[(m, a)]["gamma"], delta=PROFILES[(m, a)]["delta"]
+                    ),
+                )
+                for m in MODELS
+                for a in ACCELERATORS
+            ]
+        ),
+        service_classes=ServiceClassData(
+            spec=[
+                ServiceClassSpec(
+                    name=cls,
+                    priority=prio,
+                    model_targets=[
+                        ModelTarget(
+                            model=m, slo_itl=SLOS[(cls, m)][0], slo_ttft=SLOS[(cls, m)][1]
+                        )
+                        for m in MODELS
+                    ],
+                )
+                for cls, prio in CLASSES
+            ]
+        ),
+        servers=ServerData(spec=servers),
+        optimizer=OptimizerData(spec=OptimizerSpec(unlimited=True)),
+        capacity=CapacityData(),
+    )
+
+
+def one_cycle(rank: int, step: int, n_variants: int, device) -> dict:
+    """One full optimization cycle; returns solution stats."""
+    from wva_amd.ops import BatchedAllocationSolver
+
+    spec = build_system_spec(rank, step, n_variants)
+    system = System()
+    optimizer_spec = system.set_from_spec(spec)
+    BatchedAllocationSolver(device=device).calculate(system)
+    manager = Manager(system, Optimizer(optimizer_spec))
+    manager.optimize()
+    solution = system.generate_solution()
+
+    # SLO attainment: fraction of variants whose chosen allocation meets
+    # the class targets and is not saturated at the offered load
+    met, total = 0, 0
+    for name, server in system.servers.items():
+        total += 1
+        alloc = server.allocation
+        if alloc is None or alloc.num_replicas == 0:
+            continue
+        target = system.service_class(server.service_class_name).model_target(server.model_name)
+        if (
+            (target.itl == 0 or alloc.itl <= target.itl * 1.001)
+            and (target.ttft == 0 or alloc.ttft <= target.ttft * 1.001)
+            and not server.saturated()
+        ):
+            met += 1
+    return {"solved": len(solution.spec), "slo_met": met, "total": total}
+
+
+def main() -> None:
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=20)
+    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--variants-per-gpu", type=int, default=VARIANTS_PER_GPU)
+    args = ap.parse_args()
+
+    world_size = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    n_gpus = max(world_size, args.gpus)
+
+    has_gpu = torch.cuda.is_available()
+    if has_gpu:
+        torch.cuda.set_device(local_rank % max(torch.cuda.device_count(), 1))
+        device = "cuda"
+    else:
+        device = "cpu"
+
+    dist = None
+    if world_size > 1:
+        import torch.distributed as dist_mod
+
+        dist_mod.init_process_group(backend="nccl" if has_gpu else "gloo")
+        dist = dist_mod
+
+    def barrier_sync():
+        if dist is not None:
+            dist.barrier()
+        if has_gpu:
+            torch.cuda.synchronize()
+
+    n_variants = args.variants_per_gpu
+
+    # warmup (untimed)
+    for step in range(args.warmup):
+        one_cycle(rank, step, n_variants, device)
+
+    barrier_sync()
+    t0 = time.perf_counter()
+    stats = []
+    for step in range(args.steps):
+        stats.append(one_cycle(rank, args.warmup + step, n_variants, device))
+    barrier_sync()
+    elapsed = time.perf_counter() - t0
+
+    # max over ranks
+    if dist is not None:
+        t = torch.tensor([elapsed], dtype=torch.float64)
+        slo = torch.tensor(
+            [sum(s["slo_met"] for s in stats), sum(s["total"] for s in stats)],
+            dtype=torch.float64,
+        )
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        dist.all_reduce(slo, op=dist.ReduceOp.SUM)
+        elapsed = float(t.item())
+        slo_met, slo_total = float(slo[0].item()), float(slo[1].item())
+    else:
+        slo_met = float(sum(s["slo_met"] for s in stats))
+        slo_total = float(sum(s["total"] for s in stats))
+
+    ms_per_step = elapsed / args.steps * 1000.0
+    slo_attainment = 100.0 * slo_met / slo_total if slo_total else 0.0
+
+    if rank == 0:
+        print(
+            json.dumps(
+                {
+                    "metric": "solver_wall_clock_ms",
+                    "value": ms_per_step,
+                    "unit": "ms",
+                    "n_gpus": n_gpus,
+                    "steps": args.steps,
+                    "warmup": args.warmup,
+                    "ms_per_step": ms_per_step,
+                    "higher_is_better": False,
+                    "scaling": "weak",
+                    "vs_baseline": None,
+                    "dtype": "fp64",
+                    "data": "synthetic",
+                    "slo_attainment_pct": slo_attainment,
+                    "config": {
+                        "model": "3 models x 2 service classes (config #3)",
+                        "variants": n_gpus * n_variants,
+                        "accelerator_types": len(ACCELERATORS),
+                        "sizing_problems_per_step": n_gpus * n_variants * len(ACCELERATORS),
+                        "global_batch": n_gpus * n_variants,
+                        "seq_len": 0,
+                        "parallelism": f"fleet-shard dp{n_gpus}",
+                        "analyzer_device": device,
+                    },
+                }
+            )
+        )
+
+    if dist is not None:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()
