@@ -62,6 +62,13 @@ def binary_search(
             return x, InRegion
         y_bounds.append(y)
 
+    if within_tolerance(y_bounds[0], y_bounds[1], tolerance):
+        # flat function: direction detection would be decided by rounding
+        # noise (e.g. a batch-1 queue's ITL curve); classify by value only
+        if y_target > max(y_bounds):
+            return x_max, AboveRegion
+        return x_min, BelowRegion
+
     increasing = y_bounds[0] < y_bounds[1]
     if (increasing and y_target < y_bounds[0]) or (not increasing and y_target > y_bounds[0]):
         return x_min, BelowRegion

@@ -38,6 +38,15 @@ inline int binary_search_host(double x_min, double x_max, double y_target, F eva
     *x_star = x_max;
     return 0;
   }
+  if (within_tolerance(y0, y1, kTolerance)) {
+    // flat function: classify by value only (direction would be noise)
+    if (y_target > fmax(y0, y1)) {
+      *x_star = x_max;
+      return +1;
+    }
+    *x_star = x_min;
+    return -1;
+  }
   bool increasing = y0 < y1;
   if ((increasing && y_target < y0) || (!increasing && y_target > y0)) {
     *x_star = x_min;
