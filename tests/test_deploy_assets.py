@@ -1,0 +1,129 @@
+"""Deploy-asset consistency: YAML manifests parse and agree with the code
+(CRD schema names, ConfigMap names, sample VA round-trips through the API
+types)."""
+
+from pathlib import Path
+
+import json
+import yaml
+
+from wva_amd.api import v1alpha1
+from wva_amd.controller.reconciler import (
+    ACCELERATOR_COSTS_CM,
+    CONFIG_MAP_NAME,
+    CONFIG_MAP_NAMESPACE,
+    SERVICE_CLASSES_CM,
+)
+from wva_amd.controller.utils import create_system_data, find_model_slo
+
+DEPLOY = Path(__file__).resolve().parent.parent / "deploy"
+
+
+def load(relpath):
+    docs = list(yaml.safe_load_all((DEPLOY / relpath).read_text()))
+    return docs[0] if len(docs) == 1 else docs
+
+
+class TestCRD:
+    def test_crd_identity(self):
+        crd = load("crd/llmd.ai_variantautoscalings.yaml")
+        assert crd["spec"]["group"] == v1alpha1.GROUP
+        assert crd["spec"]["names"]["kind"] == v1alpha1.KIND
+        assert crd["spec"]["names"]["shortNames"] == [v1alpha1.SHORT_NAME]
+        ver = crd["spec"]["versions"][0]
+        assert ver["name"] == v1alpha1.VERSION
+        assert ver["subresources"] == {"status": {}}
+        cols = [(c["name"], c["jsonPath"]) for c in ver["additionalPrinterColumns"]]
+        assert cols == v1alpha1.types.PRINT_COLUMNS
+
+    def test_status_string_patterns(self):
+        crd = load("crd/llmd.ai_variantautoscalings.yaml")
+        status = crd["spec"]["versions"][0]["schema"]["openAPIV3Schema"]["properties"]["status"]
+        cur = status["properties"]["currentAlloc"]["properties"]
+        for field in ("variantCost", "itlAverage", "ttftAverage"):
+            assert cur[field]["pattern"] == r"^\d+(\.\d+)?$"
+
+
+class TestConfigMaps:
+    def test_names_match_controller(self):
+        for relpath, want in (
+            ("configmap-accelerator-unitcost.yaml", ACCELERATOR_COSTS_CM),
+            ("configmap-service-classes.yaml", SERVICE_CLASSES_CM),
+            ("configmap-controller.yaml", CONFIG_MAP_NAME),
+        ):
+            cm = load(relpath)
+            assert cm["metadata"]["name"] == want
+            assert cm["metadata"]["namespace"] == CONFIG_MAP_NAMESPACE
+
+    def test_accelerator_table_parses(self):
+        cm = load("configmap-accelerator-unitcost.yaml")
+        table = {k: json.loads(v) for k, v in cm["data"].items()}
+        assert table["MI355X"]["device"] == "AMD-MI355X-288GB"
+        assert float(table["MI355X"]["cost"]) > 0
+        # feeds the spec adapter
+        sd = create_system_data(table, {})
+        names = {a.name for a in sd.spec.accelerators.spec}
+        assert names == {"MI355X", "MI300X", "L40S"}
+        mi = next(a for a in sd.spec.accelerators.spec if a.name == "MI355X")
+        assert mi.mem_size == 288
+
+    def test_service_classes_parse_and_resolve(self):
+        cm = load("configmap-service-classes.yaml")
+        entry, cls = find_model_slo(cm["data"], "default/llama-3.1-8b")
+        assert cls == "Premium"
+        assert entry.slo_tpot == 9  # the reference demo's Premium TPOT target
+        assert entry.slo_ttft == 1000
+
+
+class TestSampleVA:
+    def test_round_trips_through_api_types(self):
+        doc = load("samples/mi355x-variantautoscaling.yaml")
+        va = v1alpha1.VariantAutoscaling.model_validate(doc)
+        assert va.spec.model_id == "default/llama-3.1-8b"
+        assert va.metadata.labels["inference.optimization/acceleratorName"] == "MI355X"
+        profile = va.spec.model_profile.accelerators[0]
+        assert profile.acc == "MI355X"
+        assert float(profile.perf_parms.decode_parms["alpha"]) > 0
+        # serialization keeps camelCase keys
+        out = va.to_dict()
+        assert out["spec"]["modelID"] == "default/llama-3.1-8b"
+        assert "maxBatchSize" in out["spec"]["modelProfile"]["accelerators"][0]
+
+
+class TestIntegrations:
+    def test_hpa_and_keda_reference_inferno_metric(self):
+        hpa = load("integrations/hpa.yaml")
+        metric = hpa["spec"]["metrics"][0]["external"]["metric"]
+        assert metric["name"] == "inferno_desired_replicas"
+        keda = load("integrations/keda-scaledobject.yaml")
+        assert keda["spec"]["minReplicaCount"] == 0  # scale-to-zero capable
+        assert "inferno_desired_replicas" in keda["spec"]["triggers"][0]["metadata"]["query"]
+
+    def test_emulator_manifest(self):
+        docs = load("emulator/vllm-emulator.yaml")
+        deploy = docs[0]
+        env = {e["name"]: e.get("value") for e in deploy["spec"]["template"]["spec"]["containers"][0]["env"]}
+        assert env["MEM_SIZE"] == "294912"  # 288 GB MI355X
+
+
+class TestMainEntry:
+    def test_parse_args_defaults(self):
+        from wva_amd.__main__ import parse_args
+
+        args = parse_args([])
+        assert args.metrics_bind_address == ":8443"
+        assert args.health_probe_bind_address == ":8081"
+        assert args.leader_elect is False
+        assert args.metrics_secure is True
+        assert args.enable_http2 is False
+
+    def test_leader_lock(self, tmp_path):
+        from wva_amd.__main__ import acquire_leader_lock
+        import os
+
+        lock_path = str(tmp_path / "leader.lock")
+        fd = acquire_leader_lock(lock_path)
+        try:
+            assert (tmp_path / "leader.lock").read_text() == str(os.getpid())
+        finally:
+            os.close(fd)
