@@ -1,0 +1,104 @@
+"""Tests for the profiling/telemetry tools: perf-parameter fitting (tiny on
+CPU, real curves on GPU) and the amd-smi exporter."""
+
+import numpy as np
+import pytest
+
+from amd_smi_exporter import AmdSmiExporter
+from profiler.fit_perf_params import _linfit, fit
+
+
+class TestLinFit:
+    def test_exact_line(self):
+        x = np.array([1.0, 2.0, 4.0, 8.0])
+        y = 3.0 + 0.5 * x
+        a, b, r2 = _linfit(x, y)
+        assert a == pytest.approx(3.0)
+        assert b == pytest.approx(0.5)
+        assert r2 == pytest.approx(1.0)
+
+    def test_noisy_line(self):
+        rng = np.random.default_rng(0)
+        x = np.linspace(1, 64, 12)
+        y = 7.0 + 0.04 * x + rng.normal(0, 0.01, size=x.size)
+        a, b, r2 = _linfit(x, y)
+        assert a == pytest.approx(7.0, abs=0.05)
+        assert b == pytest.approx(0.04, rel=0.05)
+        assert r2 > 0.99
+
+
+class TestFitCPU:
+    def test_tiny_fit_runs_on_cpu(self):
+        result = fit(
+            layers=1,
+            hidden=64,
+            heads=4,
+            batches=[1, 2],
+            seq_len=8,
+            decode_iters=2,
+            warmup=1,
+            device="cpu",
+        )
+        assert result.alpha >= 0 or result.beta != 0  # a fit was produced
+        assert len(result.decode_points) == 2
+        assert len(result.prefill_points) == 2
+        assert all(ms > 0 for _, ms in result.decode_points)
+
+
+@pytest.mark.gpu
+class TestFitGPU:
+    def test_mi355x_curves_are_linear_enough(self):
+        import torch
+
+        assert torch.cuda.is_available()
+        result = fit(
+            layers=4,
+            hidden=2048,
+            heads=16,
+            batches=[1, 4, 16, 64],
+            seq_len=256,
+            decode_iters=20,
+            warmup=5,
+        )
+        # decode time must grow with batch and fit the linear law well
+        assert result.beta > 0
+        assert result.r2_decode > 0.8
+        assert result.r2_prefill > 0.8
+        assert result.alpha > 0
+
+
+class TestAmdSmiExporter:
+    def test_synthetic_collect(self):
+        exporter = AmdSmiExporter(synthetic=True)
+        n = exporter.collect_once()
+        assert n == 1
+        v = exporter.registry.get_sample_value("amd_smi_gpu_gfx_activity", {"gpu_id": "0"})
+        assert v == 42.0
+        assert exporter.registry.get_sample_value(
+            "amd_smi_gpu_vram_used_bytes", {"gpu_id": "0"}
+        ) == 128 * 1024**3
+
+    def test_scrapeable_by_promlib(self):
+        from prometheus_client import generate_latest
+
+        from wva_amd.controller.collector import collect_gpu_telemetry
+        from wva_amd.promlib import PromlibAPI, Scraper, TimeSeriesStore
+
+        exporter = AmdSmiExporter(synthetic=True)
+        exporter.collect_once()
+        store = TimeSeriesStore()
+        scraper = Scraper(store)
+        scraper.add_target(
+            lambda: generate_latest(exporter.registry), extra_labels={"namespace": "default"}
+        )
+        scraper.scrape_once()
+        telemetry = collect_gpu_telemetry(PromlibAPI(store), "default")
+        assert telemetry is not None
+        assert telemetry.utilization_pct == 42.0
+        assert telemetry.power_watts == 750.0
+
+    @pytest.mark.gpu
+    def test_real_gpu_readout(self):
+        exporter = AmdSmiExporter()
+        n = exporter.collect_once()
+        assert n >= 1  # at least one MI355X visible

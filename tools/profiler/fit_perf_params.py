@@ -1,0 +1,227 @@
+"""Fit VariantAutoscaling perf parameters from measured MI355X curves.
+
+The autoscaler's queue analyzer is parameterized by linear latency laws
+
+    ITL  = alpha + beta  * batchSize            (decode step, ms)
+    TTFT = gamma + delta * inTokens * batchSize (prefill, ms)
+
+The reference fits these from guidellm runs against a live vLLM server
+(/root/reference/docs/tutorials/parameter-estimation.md:24-265: a
+synchronous run and a max-concurrency run, two-point fit).  This tool
+measures them directly on the accelerator: it runs a random-init
+transformer decoder (bf16, KV-cached decode / full prefill) across a
+batch-size sweep and least-squares fits the laws.  Output is a ready
+``perfParms`` block for the VA profile.
+
+Usage (on an MI355X box):
+    python tools/profiler/fit_perf_params.py --layers 8 --hidden 4096 \
+        --batches 1 2 4 8 16 32 64 --out gpurun_out/perf_parms.json
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import time
+from dataclasses import dataclass
+from typing import List, Tuple
+
+import numpy as np
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+
+class DecoderLayer(nn.Module):
+    def __init__(self, hidden: int, heads: int, ffn_mult: int = 4) -> None:
+        super().__init__()
+        self.heads = heads
+        self.head_dim = hidden // heads
+        self.qkv = nn.Linear(hidden, 3 * hidden, bias=False)
+        self.o = nn.Linear(hidden, hidden, bias=False)
+        self.up = nn.Linear(hidden, ffn_mult * hidden, bias=False)
+        self.down = nn.Linear(ffn_mult * hidden, hidden, bias=False)
+        self.norm1 = nn.LayerNorm(hidden)
+        self.norm2 = nn.LayerNorm(hidden)
+
+    def forward(self, x, kv_cache=None):
+        # x: [B, T, H]
+        B, T, H = x.shape
+        residual = x
+        x = self.norm1(x)
+        qkv = self.qkv(x).view(B, T, 3, self.heads, self.head_dim)
+        q, k, v = qkv.unbind(2)  # [B, T, heads, hd]
+        q, k, v = (t.transpose(1, 2) for t in (q, k, v))  # [B, heads, T, hd]
+        if kv_cache is not None:
+            k = torch.cat([kv_cache[0], k], dim=2)
+            v = torch.cat([kv_cache[1], v], dim=2)
+        attn = F.scaled_dot_product_attention(q, k, v, is_causal=kv_cache is None)
+        x = self.o(attn.transpose(1, 2).reshape(B, T, H))
+        x = residual + x
+        x = x + self.down(F.silu(self.up(self.norm2(x))))
+        return x, (k, v)
+
+
+class TinyDecoder(nn.Module):
+    def __init__(self, layers: int, hidden: int, heads: int, vocab: int = 32000) -> None:
+        super().__init__()
+        self.embed = nn.Embedding(vocab, hidden)
+        self.layers = nn.ModuleList(DecoderLayer(hidden, heads) for _ in range(layers))
+        self.head = nn.Linear(hidden, vocab, bias=False)
+
+    def prefill(self, tokens):
+        x = self.embed(tokens)
+        caches = []
+        for layer in self.layers:
+            x, kv = layer(x)
+            caches.append(kv)
+        return self.head(x[:, -1:]), caches
+
+    def decode_step(self, tokens, caches):
+        x = self.embed(tokens)
+        new_caches = []
+        for layer, kv in zip(self.layers, caches):
+            x, new_kv = layer(x, kv_cache=kv)
+            new_caches.append(new_kv)
+        return self.head(x), new_caches
+
+
+@dataclass
+class FitResult:
+    alpha: float
+    beta: float
+    gamma: float
+    delta: float
+    decode_points: List[Tuple[int, float]]
+    prefill_points: List[Tuple[int, float]]
+    r2_decode: float
+    r2_prefill: float
+
+
+def _linfit(x: np.ndarray, y: np.ndarray) -> Tuple[float, float, float]:
+    A = np.stack([np.ones_like(x), x], axis=1)
+    coef, *_ = np.linalg.lstsq(A, y, rcond=None)
+    pred = A @ coef
+    ss_res = float(((y - pred) ** 2).sum())
+    ss_tot = float(((y - y.mean()) ** 2).sum())
+    r2 = 1.0 - ss_res / ss_tot if ss_tot > 0 else 1.0
+    return float(coef[0]), float(coef[1]), r2
+
+
+def measure(
+    model: TinyDecoder,
+    device: str,
+    batches: List[int],
+    seq_len: int,
+    decode_iters: int,
+    warmup: int,
+) -> FitResult:
+    def sync():
+        if device.startswith("cuda"):
+            torch.cuda.synchronize()
+
+    decode_points = []
+    prefill_points = []
+    with torch.no_grad():
+        for B in batches:
+            tokens = torch.randint(0, 31999, (B, seq_len), device=device)
+            # prefill timing
+            for _ in range(warmup):
+                _, caches = model.prefill(tokens)
+            sync()
+            t0 = time.perf_counter()
+            for _ in range(max(decode_iters // 4, 1)):
+                _, caches = model.prefill(tokens)
+            sync()
+            prefill_ms = (time.perf_counter() - t0) / max(decode_iters // 4, 1) * 1000.0
+            prefill_points.append((B * seq_len, prefill_ms))
+
+            # decode timing with KV cache
+            step_tokens = torch.randint(0, 31999, (B, 1), device=device)
+            for _ in range(warmup):
+                model.decode_step(step_tokens, caches)
+            sync()
+            t0 = time.perf_counter()
+            for _ in range(decode_iters):
+                model.decode_step(step_tokens, caches)
+            sync()
+            decode_ms = (time.perf_counter() - t0) / decode_iters * 1000.0
+            decode_points.append((B, decode_ms))
+
+    bx = np.array([p[0] for p in decode_points], dtype=np.float64)
+    by = np.array([p[1] for p in decode_points], dtype=np.float64)
+    alpha, beta, r2d = _linfit(bx, by)
+    px = np.array([p[0] for p in prefill_points], dtype=np.float64)
+    py = np.array([p[1] for p in prefill_points], dtype=np.float64)
+    gamma, delta, r2p = _linfit(px, py)
+    return FitResult(alpha, beta, gamma, delta, decode_points, prefill_points, r2d, r2p)
+
+
+def fit(
+    *,
+    layers: int = 8,
+    hidden: int = 2048,
+    heads: int = 16,
+    batches: List[int] = (1, 2, 4, 8, 16, 32),
+    seq_len: int = 256,
+    decode_iters: int = 20,
+    warmup: int = 3,
+    device: str = None,
+    dtype=torch.bfloat16,
+) -> FitResult:
+    device = device or ("cuda" if torch.cuda.is_available() else "cpu")
+    if device == "cpu":
+        dtype = torch.float32
+    model = TinyDecoder(layers, hidden, heads).to(device=device, dtype=dtype).eval()
+    return fit_with_model(model, device, list(batches), seq_len, decode_iters, warmup)
+
+
+def fit_with_model(model, device, batches, seq_len, decode_iters, warmup) -> FitResult:
+    return measure(model, device, batches, seq_len, decode_iters, warmup)
+
+
+def main() -> None:
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--layers", type=int, default=8)
+    ap.add_argument("--hidden", type=int, default=4096)
+    ap.add_argument("--heads", type=int, default=32)
+    ap.add_argument("--batches", type=int, nargs="+", default=[1, 2, 4, 8, 16, 32, 64])
+    ap.add_argument("--seq-len", type=int, default=512)
+    ap.add_argument("--decode-iters", type=int, default=40)
+    ap.add_argument("--warmup", type=int, default=5)
+    ap.add_argument("--acc", default="MI355X")
+    ap.add_argument("--out", default="")
+    args = ap.parse_args()
+
+    result = fit(
+        layers=args.layers,
+        hidden=args.hidden,
+        heads=args.heads,
+        batches=args.batches,
+        seq_len=args.seq_len,
+        decode_iters=args.decode_iters,
+        warmup=args.warmup,
+    )
+    payload = {
+        "acc": args.acc,
+        "model": f"tiny-decoder-L{args.layers}-H{args.hidden}",
+        "perfParms": {
+            "decodeParms": {"alpha": f"{result.alpha:.4f}", "beta": f"{result.beta:.6f}"},
+            "prefillParms": {"gamma": f"{result.gamma:.4f}", "delta": f"{result.delta:.8f}"},
+        },
+        "fit": {
+            "r2_decode": result.r2_decode,
+            "r2_prefill": result.r2_prefill,
+            "decode_points_ms": result.decode_points,
+            "prefill_points_ms": result.prefill_points,
+        },
+    }
+    text = json.dumps(payload, indent=2)
+    print(text)
+    if args.out:
+        with open(args.out, "w") as f:
+            f.write(text)
+
+
+if __name__ == "__main__":
+    main()
