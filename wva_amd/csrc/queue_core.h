@@ -106,16 +106,36 @@ struct Stats {
   double n_serv;      // avg requests in service
 };
 
+// Mode of the log-probability curve logp[n] = n*log(lam) - cum[n-1].
+// The increment logp[n+1]-logp[n] = log(lam) - log(mu(n)) is nonincreasing
+// in n because mu(n) = serv_rate(min(n+1, N)) is nondecreasing, so logp is
+// concave and its maximum sits where the increment first turns <= 0:
+// an O(log N) search over the monotone service-rate curve instead of an
+// O(K) max sweep.  Returns the state index n* in [0, K].
+WVA_HD int log_mode_state(const Parms &p, int K, double lam) {
+  if (lam >= serv_rate(p, p.max_batch)) return K;  // all increments > 0
+  if (lam <= serv_rate(p, 1)) return 0;
+  // smallest b in [1, N] with serv_rate(b) >= lam; mode = b - 1
+  int lo = 1, hi = p.max_batch;
+  while (lo < hi) {
+    int mid = (lo + hi) / 2;
+    if (serv_rate(p, mid) < lam) {
+      lo = mid + 1;
+    } else {
+      hi = mid;
+    }
+  }
+  return lo - 1;
+}
+
 // Scalar model evaluation at arrival rate lam (req/ms) given the inclusive
 // cumulative sum cum[n] = sum_{i<=n} log_mu(i), n in [0, K-1].
 WVA_HD Stats eval_model(const Parms &p, const double *cum, int K, double lam) {
   double loglam = log(lam);
-  // max of logp over n = 0..K (logp(0) = 0)
-  double m = 0.0;
-  for (int n = 1; n <= K; ++n) {
-    double lp = n * loglam - cum[n - 1];
-    if (lp > m) m = lp;
-  }
+  // max of logp over n = 0..K via the concavity closed form
+  int n_star = log_mode_state(p, K, lam);
+  double m = (n_star == 0) ? 0.0 : n_star * loglam - cum[n_star - 1];
+  if (m < 0.0) m = 0.0;  // logp(0) = 0 participates in the max
   int num = p.max_batch;  // serv_rate array length
   double S = 0.0, Ni = 0.0, Snum = 0.0, Ninum = 0.0, eK = 0.0;
   for (int n = 0; n <= K; ++n) {

@@ -51,21 +51,14 @@ struct WgEval {
     const int lane = tid & 63;
     const double loglam = log(lam);
 
-    // pass 1: max of logp over n = 0..K (logp(0) = 0 handled by init)
-    double lmax = (tid == 0) ? 0.0 : -INFINITY;
-    for (int n = tid + 1; n <= K; n += WVA_THREADS) {
-      double lp = (double)n * loglam - cum[n - 1];
-      lmax = fmax(lmax, lp);
-    }
-    for (int off = 32; off > 0; off >>= 1)
-      lmax = fmax(lmax, __shfl_down(lmax, off, 64));
+    // max of logp via the concavity closed form (queue_core.h): every
+    // thread computes the identical O(log N) search — no sweep, no barrier
+    const int n_star = log_mode_state(p, K, lam);
+    double m = (n_star == 0) ? 0.0 : (double)n_star * loglam - cum[n_star - 1];
+    if (m < 0.0) m = 0.0;
     __syncthreads();  // red may still be read from a previous eval
-    if (lane == 0) red[17 + wave] = lmax;
-    __syncthreads();
-    double m = red[17];
-    for (int w = 1; w < WVA_WAVES; ++w) m = fmax(m, red[17 + w]);
 
-    // pass 2: normalization and moment sums
+    // single pass: normalization and moment sums
     const int num = p.max_batch;
     double S = 0.0, Ni = 0.0, Snum = 0.0, Ninum = 0.0;
     if (tid == 0) {
