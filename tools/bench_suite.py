@@ -1,0 +1,257 @@
+"""Benchmark harness for the five BASELINE.json configurations.
+
+Each config reports the headline pair: solver wall-clock (ms per global
+solve) and SLO-attainment (%).  Configs #1/#5 run the full control loop
+against the in-process emulator stack; #2/#3/#4 run the optimization path
+on synthetic fleets of the named shape.
+
+    python tools/bench_suite.py [--configs 1 2 3 4 5] [--out results.json]
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import sys
+import time
+from pathlib import Path
+
+ROOT = Path(__file__).resolve().parent.parent
+sys.path.insert(0, str(ROOT))
+sys.path.insert(0, str(ROOT / "tools"))
+sys.path.insert(0, str(ROOT / "tests"))
+
+import numpy as np
+
+
+def _timed_cycles(run_cycle, steps=10, warmup=3):
+    for _ in range(warmup):
+        run_cycle()
+    t0 = time.perf_counter()
+    stats = [run_cycle() for _ in range(steps)]
+    elapsed = time.perf_counter() - t0
+    met = sum(s[0] for s in stats)
+    total = sum(s[1] for s in stats)
+    return {
+        "solver_wall_clock_ms": elapsed / steps * 1000.0,
+        "slo_attainment_pct": 100.0 * met / total if total else 0.0,
+        "steps": steps,
+    }
+
+
+def _fleet_cycle_factory(n_variants, device, step_holder):
+    import bench
+
+    def run_cycle():
+        step_holder[0] += 1
+        s = bench.one_cycle(0, step_holder[0], n_variants, device)
+        return s["slo_met"], s["total"]
+
+    return run_cycle
+
+
+def config_1():
+    """1 VA, 1 model, emulator + fake cluster (CPU-only plumbing)."""
+    from prometheus_client import CollectorRegistry
+
+    from wva_amd.controller import metrics as ctrl_metrics
+    from wva_amd.controller.promclient import MockPromAPI
+    from wva_amd.controller.reconciler import VariantAutoscalingReconciler
+    from wva_amd.api import v1alpha1
+    from kube_fixtures import make_cluster, make_deployment, make_va, set_load_metrics
+
+    registry = CollectorRegistry()
+    ctrl_metrics.init_metrics(registry)
+    try:
+        cluster = make_cluster()
+        make_deployment(cluster, replicas=1)
+        make_va(cluster)
+        prom = MockPromAPI()
+        set_load_metrics(prom, "default/llama-8b", "default", arrival_rps=5.0)
+        rec = VariantAutoscalingReconciler(cluster, prom)
+
+        def run_cycle():
+            rec.reconcile()
+            va = cluster.get(v1alpha1.VariantAutoscaling, "vllm-llama", "default")
+            ok = va.status.desired_optimized_alloc.num_replicas >= 1
+            return (1 if ok else 0), 1
+
+        return _timed_cycles(run_cycle)
+    finally:
+        ctrl_metrics.reset_metrics()
+
+
+def config_2(device):
+    """Llama-3.1-8B Premium on 1x MI355X (measured perf parms), analyzer fed
+    by a synthetic live load."""
+    from wva_amd.config import (
+        AllocationData,
+        ServerLoadSpec,
+        ServerSpec,
+    )
+    import bench
+    from wva_amd.core import System
+    from wva_amd.solver import Manager, Optimizer
+    from wva_amd.ops import BatchedAllocationSolver
+
+    step = [0]
+
+    def run_cycle():
+        step[0] += 1
+        rng = np.random.default_rng(step[0])
+        spec = bench.build_system_spec(0, step[0], 1)
+        # narrow to the single-VA MI355X shape with the measured profile
+        spec.servers.spec = [
+            ServerSpec(
+                name="llama-8b:prod",
+                class_name="Premium",
+                model="llama-8b",
+                min_num_replicas=1,
+                max_batch_size=512,
+                current_alloc=AllocationData(
+                    accelerator="MI355X",
+                    num_replicas=1,
+                    load=ServerLoadSpec(
+                        arrival_rate=float(rng.uniform(60, 1200)),
+                        avg_in_tokens=512,
+                        avg_out_tokens=128,
+                    ),
+                ),
+            )
+        ]
+        for pd in spec.models.perf_data:
+            if pd.name == "llama-8b" and pd.acc == "MI355X":
+                # MI355X-measured llama-8B-scale parameters
+                pd.decode_parms.alpha, pd.decode_parms.beta = 4.95, 0.369
+                pd.prefill_parms.gamma, pd.prefill_parms.delta = 3.02, 0.0101 / 512
+        system = System()
+        opt_spec = system.set_from_spec(spec)
+        BatchedAllocationSolver(device=device).calculate(system)
+        Manager(system, Optimizer(opt_spec)).optimize()
+        server = system.server("llama-8b:prod")
+        alloc = server.allocation
+        target = system.service_class("Premium").model_target("llama-8b")
+        ok = (
+            alloc is not None
+            and alloc.itl <= target.itl * 1.001
+            and alloc.ttft <= target.ttft * 1.001
+            and not server.saturated()
+        )
+        return (1 if ok else 0), 1
+
+    return _timed_cycles(run_cycle)
+
+
+def config_3(device):
+    """3 models x 2 service classes on one 8x MI355X node (24 variants)."""
+    return _timed_cycles(_fleet_cycle_factory(24, device, [0]))
+
+
+def config_4(device):
+    """Heterogeneous pool trade-off solve at fleet scale (64 variants x 3
+    accelerator types, candidates unpinned)."""
+    return _timed_cycles(_fleet_cycle_factory(64, device, [100]))
+
+
+def config_5():
+    """KEDA-actuated 0->peak->0 ramp (full emulator closed loop); reports
+    the ramp trajectory plus per-cycle solve time."""
+    import os
+
+    os.environ.setdefault("WVA_RATE_WINDOW", "8s")
+    os.environ["WVA_SCALE_TO_ZERO"] = "true"
+    from prometheus_client import CollectorRegistry
+
+    from wva_amd.controller import metrics as ctrl_metrics
+    from wva_amd.controller.reconciler import VariantAutoscalingReconciler
+    from wva_amd.promlib import PromlibAPI, Scraper, TimeSeriesStore
+    from vllm_emulator.engine import EmulatorSettings
+    from kube_fixtures import make_cluster, make_deployment, make_va
+    from test_e2e import EmulatorProcess, drive_load, desired_replicas, simulate_hpa
+
+    registry = CollectorRegistry()
+    ctrl_metrics.init_metrics(registry)
+    try:
+        settings = EmulatorSettings(
+            model="default/llama-8b",
+            decode_alpha=4.0,
+            decode_beta=0.05,
+            prefill_gamma=4.0,
+            prefill_delta=0.01,
+            avg_generated_len=30,
+            tokens_distribution="deterministic",
+            max_batch_size=8,
+            realtime=True,
+        )
+        cluster = make_cluster(opt_interval="1s")
+        make_deployment(cluster, replicas=0)
+        make_va(cluster, max_batch=8, alpha="4.0", beta="0.05", gamma="4.0", delta="0.01")
+        store = TimeSeriesStore()
+        scraper = Scraper(store)
+        rec = VariantAutoscalingReconciler(cluster, PromlibAPI(store))
+        trajectory = []
+        cycle_ms = []
+        with EmulatorProcess(settings) as emu:
+            scraper.add_target(f"{emu.base_url}/metrics", extra_labels={"namespace": "default"})
+            scraper.start(interval=0.5)
+            try:
+                for phase in ("idle", "peak", "idle2"):
+                    if phase == "peak":
+                        drive_load(emu.base_url, rate_rps=30.0, duration_s=6.0)
+                    else:
+                        time.sleep(10.0 if phase == "idle2" else 2.0)
+                    t0 = time.perf_counter()
+                    rec.reconcile()
+                    cycle_ms.append((time.perf_counter() - t0) * 1000.0)
+                    trajectory.append(desired_replicas(cluster))
+                    simulate_hpa(cluster, registry)
+            finally:
+                scraper.stop()
+        ramp_ok = trajectory[0] == 0 and trajectory[1] >= 1 and trajectory[2] == 0
+        return {
+            "solver_wall_clock_ms": float(np.mean(cycle_ms)),
+            "slo_attainment_pct": 100.0 if ramp_ok else 0.0,
+            "trajectory": trajectory,
+            "steps": len(cycle_ms),
+        }
+    finally:
+        ctrl_metrics.reset_metrics()
+
+
+def main() -> None:
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--configs", type=int, nargs="+", default=[1, 2, 3, 4, 5])
+    ap.add_argument("--out", default="")
+    args = ap.parse_args()
+
+    import logging
+
+    from wva_amd.controller.logger import log as wva_log
+
+    wva_log.init(logging.ERROR)  # keep the harness output to JSON lines
+
+    import torch
+
+    device = "cuda" if torch.cuda.is_available() else "cpu"
+    runners = {
+        1: ("1 VA + emulator plumbing (CPU control loop)", config_1),
+        2: ("Llama-3.1-8B Premium on 1x MI355X (measured parms)", lambda: config_2(device)),
+        3: ("3 models x 2 classes, 8x MI355X node", lambda: config_3(device)),
+        4: ("heterogeneous MI355X+MI300X+L40S pool, 64 variants", lambda: config_4(device)),
+        5: ("KEDA 0->peak->0 ramp (closed loop)", config_5),
+    }
+    results = {}
+    for c in args.configs:
+        name, fn = runners[c]
+        res = fn()
+        res["name"] = name
+        res["analyzer_device"] = device if c in (2, 3, 4) else "cpu"
+        results[f"config_{c}"] = res
+        print(json.dumps({f"config_{c}": res}))
+    if args.out:
+        with open(args.out, "w") as f:
+            json.dump(results, f, indent=2)
+
+
+if __name__ == "__main__":
+    main()
