@@ -128,19 +128,58 @@ WVA_HD int log_mode_state(const Parms &p, int K, double lam) {
   return lo - 1;
 }
 
+WVA_HD double log_p(const double *cum, double loglam, int n) {
+  return (n == 0) ? 0.0 : n * loglam - cum[n - 1];
+}
+
+// States more than kLogCutoff nats below the mode contribute < ~1e-16 to
+// any normalized sum (at most K * e^-45 relative mass); since logp is
+// concave, the significant window [lo, hi] around the mode is found by
+// two O(log K) bisections instead of sweeping all K+1 states.
+constexpr double kLogCutoff = 45.0;
+
+WVA_HD void state_window(const double *cum, double loglam, int K, int n_star,
+                         double m, int *lo_out, int *hi_out) {
+  const double thresh = m - kLogCutoff;
+  // left edge: logp nondecreasing on [0, n_star]; first n with logp >= thresh
+  int lo = 0, hi = n_star;
+  while (lo < hi) {
+    int mid = (lo + hi) / 2;
+    if (log_p(cum, loglam, mid) < thresh) {
+      lo = mid + 1;
+    } else {
+      hi = mid;
+    }
+  }
+  *lo_out = lo;
+  // right edge: logp nonincreasing on [n_star, K]; last n with logp >= thresh
+  lo = n_star;
+  hi = K;
+  while (lo < hi) {
+    int mid = (lo + hi + 1) / 2;
+    if (log_p(cum, loglam, mid) < thresh) {
+      hi = mid - 1;
+    } else {
+      lo = mid;
+    }
+  }
+  *hi_out = lo;
+}
+
 // Scalar model evaluation at arrival rate lam (req/ms) given the inclusive
 // cumulative sum cum[n] = sum_{i<=n} log_mu(i), n in [0, K-1].
 WVA_HD Stats eval_model(const Parms &p, const double *cum, int K, double lam) {
   double loglam = log(lam);
   // max of logp over n = 0..K via the concavity closed form
   int n_star = log_mode_state(p, K, lam);
-  double m = (n_star == 0) ? 0.0 : n_star * loglam - cum[n_star - 1];
+  double m = log_p(cum, loglam, n_star);
   if (m < 0.0) m = 0.0;  // logp(0) = 0 participates in the max
+  int n_lo, n_hi;
+  state_window(cum, loglam, K, n_star, m, &n_lo, &n_hi);
   int num = p.max_batch;  // serv_rate array length
   double S = 0.0, Ni = 0.0, Snum = 0.0, Ninum = 0.0, eK = 0.0;
-  for (int n = 0; n <= K; ++n) {
-    double lp = (n == 0) ? 0.0 : n * loglam - cum[n - 1];
-    double e = exp(lp - m);
+  for (int n = n_lo; n <= n_hi; ++n) {
+    double e = exp(log_p(cum, loglam, n) - m);
     S += e;
     Ni += n * e;
     if (n <= num) {

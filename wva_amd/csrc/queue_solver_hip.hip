@@ -8,8 +8,10 @@
 //   - the cumulative log-service-rate table (K = 11*N doubles) lives in LDS
 //     and is built with a chunked parallel scan — every later model
 //     evaluation is LDS-bandwidth bound, never HBM;
-//   - each model evaluation is two strided passes over the K+1 states with
-//     wave64 __shfl_down reductions (max, then 4 sums + boundary term);
+//   - each model evaluation is ONE strided pass over the significant
+//     state window (the normalization max and the ~1e-16 cutoff window
+//     both come from closed-form O(log K) searches over the concave
+//     log-probability curve) with wave64 __shfl_down reductions;
 //     consecutive lanes touch consecutive doubles -> conflict-free
 //     ds_read_b64 (bank = (a/4) % 64);
 //   - the bisection control flow is uniform across the workgroup (all
@@ -35,9 +37,7 @@ namespace wva {
 #define WVA_WAVES (WVA_THREADS / 64)
 
 // Reduction scratch layout (doubles, after cum[max_k] in dynamic LDS):
-//   red[0..WVA_WAVES*4-1]  per-wave partial sums (4 values at once)
-//   red[16]                boundary term exp(logp(K) - m)
-//   red[17..17+WVA_WAVES]  per-wave max partials
+//   red[0..WVA_WAVES*5-1]  per-wave partials (S, Ni, Snum, Ninum, eK)
 
 struct WgEval {
   const Parms &p;
@@ -51,54 +51,55 @@ struct WgEval {
     const int lane = tid & 63;
     const double loglam = log(lam);
 
-    // max of logp via the concavity closed form (queue_core.h): every
-    // thread computes the identical O(log N) search — no sweep, no barrier
+    // max of logp via the concavity closed form, and the significant
+    // state window via two bisections (queue_core.h): every thread
+    // computes the identical O(log K) searches — no sweep, no barrier,
+    // and the exp sweep shrinks to the ~window where p(n) > 1e-16
     const int n_star = log_mode_state(p, K, lam);
-    double m = (n_star == 0) ? 0.0 : (double)n_star * loglam - cum[n_star - 1];
+    double m = log_p(cum, loglam, n_star);
     if (m < 0.0) m = 0.0;
+    int n_lo, n_hi;
+    state_window(cum, loglam, K, n_star, m, &n_lo, &n_hi);
     __syncthreads();  // red may still be read from a previous eval
 
-    // single pass: normalization and moment sums
+    // single windowed pass: normalization and moment sums (eK is the
+    // state-K boundary term, reduced like the sums — only its owner
+    // thread contributes a non-zero partial)
     const int num = p.max_batch;
-    double S = 0.0, Ni = 0.0, Snum = 0.0, Ninum = 0.0;
-    if (tid == 0) {
-      double e0 = exp(0.0 - m);  // state 0
-      S += e0;
-      Snum += e0;
-      if (threadIdx.x == 0 && 0 == K) red[16] = e0;
-    }
-    for (int n = tid + 1; n <= K; n += WVA_THREADS) {
-      double lp = (double)n * loglam - cum[n - 1];
-      double e = exp(lp - m);
+    double S = 0.0, Ni = 0.0, Snum = 0.0, Ninum = 0.0, eK = 0.0;
+    for (int n = n_lo + tid; n <= n_hi; n += WVA_THREADS) {
+      double e = exp(log_p(cum, loglam, n) - m);
       S += e;
       Ni += (double)n * e;
       if (n <= num) {
         Snum += e;
         Ninum += (double)n * e;
       }
-      if (n == K) red[16] = e;  // exactly one thread owns state K
+      if (n == K) eK = e;
     }
     for (int off = 32; off > 0; off >>= 1) {
       S += __shfl_down(S, off, 64);
       Ni += __shfl_down(Ni, off, 64);
       Snum += __shfl_down(Snum, off, 64);
       Ninum += __shfl_down(Ninum, off, 64);
+      eK += __shfl_down(eK, off, 64);
     }
     if (lane == 0) {
-      red[wave * 4 + 0] = S;
-      red[wave * 4 + 1] = Ni;
-      red[wave * 4 + 2] = Snum;
-      red[wave * 4 + 3] = Ninum;
+      red[wave * 5 + 0] = S;
+      red[wave * 5 + 1] = Ni;
+      red[wave * 5 + 2] = Snum;
+      red[wave * 5 + 3] = Ninum;
+      red[wave * 5 + 4] = eK;
     }
     __syncthreads();
-    S = Ni = Snum = Ninum = 0.0;
+    S = Ni = Snum = Ninum = eK = 0.0;
     for (int w = 0; w < WVA_WAVES; ++w) {
-      S += red[w * 4 + 0];
-      Ni += red[w * 4 + 1];
-      Snum += red[w * 4 + 2];
-      Ninum += red[w * 4 + 3];
+      S += red[w * 5 + 0];
+      Ni += red[w * 5 + 1];
+      Snum += red[w * 5 + 2];
+      Ninum += red[w * 5 + 3];
+      eK += red[w * 5 + 4];
     }
-    const double eK = red[16];
 
     Stats st;
     st.throughput = lam * (1.0 - eK / S);
