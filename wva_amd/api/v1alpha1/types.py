@@ -63,7 +63,9 @@ class ModelProfile(_Base):
 
 
 class VariantAutoscalingSpec(_Base):
-    model_id: str = Field(alias="modelID", min_length=1, default="")
+    # modelID minLength=1 is enforced by the CRD schema (deploy/crd/);
+    # the reconciler skips empty modelIDs at runtime like the reference
+    model_id: str = Field(alias="modelID", default="")
     slo_class_ref: ConfigMapKeyRef = Field(
         alias="sloClassRef", default_factory=lambda: ConfigMapKeyRef(name="x", key="x")
     )
