@@ -172,3 +172,46 @@ class TestHTTPServer:
         finally:
             server.should_exit = True
             thread.join(timeout=5.0)
+
+
+class TestWaitingQueuePolicy:
+    def test_sorted_by_token_len(self):
+        engine, _ = make_engine(max_batch_size=1, waiting_queue_policy="sorted_by_token_len")
+
+        async def scenario():
+            engine.submit(RequestElement("running", 5, 500))  # occupies the slot
+            engine.submit(RequestElement("long", 100, 200))
+            engine.submit(RequestElement("short", 10, 20))
+            engine.submit(RequestElement("mid", 50, 60))
+            assert [r.req_id for r in engine.waiting] == ["short", "mid", "long"]
+
+        run(scenario())
+
+    def test_fifo_default(self):
+        engine, _ = make_engine(max_batch_size=1)
+
+        async def scenario():
+            engine.submit(RequestElement("running", 5, 500))
+            engine.submit(RequestElement("long", 100, 200))
+            engine.submit(RequestElement("short", 10, 20))
+            assert [r.req_id for r in engine.waiting] == ["long", "short"]
+
+        run(scenario())
+
+
+class TestExperiment:
+    def test_virtual_time_simulation(self):
+        import asyncio as aio
+
+        from vllm_emulator.experiment import simulate
+
+        settings = EmulatorSettings(
+            model="exp", decode_alpha=5.0, decode_beta=0.1, prefill_gamma=5.0,
+            prefill_delta=0.01, avg_generated_len=20, max_batch_size=16, realtime=False,
+        )
+        result = aio.new_event_loop().run_until_complete(
+            simulate(settings, rate_rps=10.0, duration_s=5.0, in_tokens=32)
+        )
+        assert result["completed"] == result["submitted"] > 10
+        assert result["ttft_ms"]["mean"] > 0
+        assert result["latency_ms"]["mean"] > result["ttft_ms"]["mean"]

@@ -42,6 +42,9 @@ class EmulatorSettings:
     avg_generated_len: int = int(os.getenv("AVG_TOKENS", "100"))
     tokens_distribution: str = os.getenv("TOKENS_DISTRIBUTION", "uniform")
     realtime: bool = os.getenv("REALTIME", "true").lower() == "true"
+    # waiting-queue admission policy: "fifo" or "sorted_by_token_len"
+    # (shortest sequences first — the reference's vLLM_varitaion_sorted_wq)
+    waiting_queue_policy: str = os.getenv("WAITING_QUEUE_POLICY", "fifo")
 
 
 class Clock:
@@ -170,6 +173,8 @@ class EmulatedVLLM:
         req.entered_waiting_ms = self.clock.now_ms
         req.stage = "waiting"
         self.waiting.append(req)
+        if self.settings.waiting_queue_policy == "sorted_by_token_len":
+            self.waiting.sort(key=lambda r: r.token_len)
         self.metrics.l(self.metrics.waiting).inc()
 
     def _remove_running(self, req: RequestElement) -> None:
