@@ -172,3 +172,31 @@ class TestSystem:
         assert system.accelerator("L40S") is None
         with pytest.raises(KeyError):
             system.remove_accelerator("L40S")
+
+
+class TestScaleAndReallocate:
+    def test_scale_tracks_load_change(self):
+        from wva_amd.core.allocation import scale_allocation
+
+        system, _ = make_system(servers=[server_spec("s:ns", arrival_rate=600.0)])
+        alloc = create_allocation(system, "s:ns", "MI355X")
+        system.server("s:ns").load.arrival_rate = 6000.0
+        new_alloc, inc = scale_allocation(system, alloc, "s:ns")
+        assert new_alloc is not None
+        assert inc == new_alloc.num_replicas - alloc.num_replicas > 0
+
+    def test_reallocate_picks_min_value(self):
+        from wva_amd.core.allocation import reallocate
+
+        system, _ = make_system(servers=[server_spec("s:ns", arrival_rate=30.0)])
+        alloc, acc = reallocate(system, "s:ns")
+        assert alloc is not None
+        # cheapest feasible accelerator wins (fresh server: value == cost)
+        assert acc == "MI300X"
+
+    def test_reallocate_infeasible(self):
+        from wva_amd.core.allocation import reallocate
+
+        system, _ = make_system()
+        alloc, acc = reallocate(system, "missing")
+        assert alloc is None and acc == ""

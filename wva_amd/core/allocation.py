@@ -19,7 +19,7 @@ de-facto contract quirks:
 from __future__ import annotations
 
 import math
-from typing import TYPE_CHECKING, Optional
+from typing import TYPE_CHECKING, Optional, Tuple
 
 from ..analyzer import (
     AnalyzerError,
@@ -215,6 +215,39 @@ def create_allocation(system: "System", server_name: str, acc_name: str) -> Opti
     )
     alloc.set_value(alloc.cost)
     return alloc
+
+
+def scale_allocation(
+    system: "System", alloc: Allocation, server_name: str
+) -> Tuple[Optional[Allocation], int]:
+    """Re-size this allocation for the server's current load; returns the
+    new allocation and the replica increment (allocation.go:166-190)."""
+    server = system.server(server_name)
+    if server is None or server.load is None:
+        return None, 0
+    if system.accelerator(alloc.accelerator) is None:
+        return None, 0
+    new_alloc = create_allocation(system, server_name, alloc.accelerator)
+    if new_alloc is None:
+        return None, 0
+    return new_alloc, new_alloc.num_replicas - alloc.num_replicas
+
+
+def reallocate(system: "System", server_name: str) -> Tuple[Optional[Allocation], str]:
+    """Pick the min-value feasible allocation across all accelerators
+    (allocation.go:192-207)."""
+    min_val = 0.0
+    min_alloc: Optional[Allocation] = None
+    for acc_name in system.accelerators:
+        alloc = create_allocation(system, server_name, acc_name)
+        # quirk preserved from allocation.go:197: a zero min value keeps
+        # accepting replacements
+        if alloc is not None and (min_alloc is None or min_val == 0 or alloc.value < min_val):
+            min_val = alloc.value
+            min_alloc = alloc
+    if min_alloc is None:
+        return None, ""
+    return min_alloc, min_alloc.accelerator
 
 
 def _zero_load_allocation(server, model, acc, perf) -> Allocation:
