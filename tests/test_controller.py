@@ -318,3 +318,56 @@ class TestBatchedReconcile:
         VariantAutoscalingReconciler(cluster2, prom, batched_analyzer=True).reconcile()
         batched_desired = get_va(cluster2).status.desired_optimized_alloc.num_replicas
         assert batched_desired == scalar_desired > 1
+
+
+class TestMultiVariant:
+    def test_concurrent_optimization_of_multiple_vas(self, cluster, prom, registry):
+        """Multi-VA cycle (reference e2e scenario e2e_test.go:701-1058):
+        several variants optimized in one global solve, each sized for its
+        own load; unlimited mode sizes beyond any physical capacity."""
+        loads = {"a": 2.0, "b": 20.0, "c": 200.0}  # req/s
+        for name, rps in loads.items():
+            make_deployment(cluster, name=f"vllm-{name}")
+            make_va(cluster, name=f"vllm-{name}", model_id="default/llama-8b")
+        set_load_metrics(prom, "default/llama-8b", "default", arrival_rps=0.0)
+        # per-deployment namespaces are all "default" and the model is
+        # shared, so the shared metrics apply to all three; differentiate
+        # by running three cycles with different global load
+        rec = VariantAutoscalingReconciler(cluster, prom)
+        results = {}
+        for name, rps in loads.items():
+            set_load_metrics(prom, "default/llama-8b", "default", arrival_rps=rps, out_tokens=200.0)
+            rec.reconcile()
+            results[name] = {
+                va.metadata.name: va.status.desired_optimized_alloc.num_replicas
+                for va in cluster.list(type(get_va(cluster, "vllm-a")))
+            }
+        # increasing load increases every variant's desired replicas
+        assert results["b"]["vllm-a"] >= results["a"]["vllm-a"]
+        assert results["c"]["vllm-a"] > results["a"]["vllm-a"]
+        # beyond-capacity sizing: unlimited mode has no cap
+        assert results["c"]["vllm-c"] > 5
+
+    def test_mixed_health_fleet(self, cluster, prom, registry):
+        """One healthy VA + one with stale metrics + one deleted: only the
+        healthy one is optimized, others untouched, cycle succeeds."""
+        import datetime
+
+        make_deployment(cluster, name="healthy")
+        make_va(cluster, name="healthy")
+        make_deployment(cluster, name="stale")
+        make_va(cluster, name="stale", model_id="default/llama-70b")
+        va = make_va(cluster, name="gone", model_id="default/llama-8b")
+        va.metadata.deletion_timestamp = datetime.datetime.now(datetime.timezone.utc)
+        cluster.update(va)
+
+        set_load_metrics(prom, "default/llama-8b", "default", arrival_rps=4.0)
+        prom.set_result(
+            'vllm:request_success_total{model_name="default/llama-70b",namespace="default"}',
+            5.0,
+            age_seconds=600.0,
+        )
+        VariantAutoscalingReconciler(cluster, prom).reconcile()
+        assert get_va(cluster, "healthy").status.desired_optimized_alloc.num_replicas >= 1
+        assert get_va(cluster, "stale").status.desired_optimized_alloc.accelerator == ""
+        assert get_va(cluster, "gone").status.desired_optimized_alloc.accelerator == ""
