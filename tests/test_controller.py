@@ -371,3 +371,48 @@ class TestMultiVariant:
         assert get_va(cluster, "healthy").status.desired_optimized_alloc.num_replicas >= 1
         assert get_va(cluster, "stale").status.desired_optimized_alloc.accelerator == ""
         assert get_va(cluster, "gone").status.desired_optimized_alloc.accelerator == ""
+
+
+class TestLimitedMode:
+    def _enable_limited(self, cluster, policy="None", capacity=1):
+        import json as _json
+
+        from wva_amd.controller.reconciler import (
+            ACCELERATOR_COSTS_CM,
+            CONFIG_MAP_NAME,
+            CONFIG_MAP_NAMESPACE,
+        )
+
+        cm = cluster.get(ConfigMap, CONFIG_MAP_NAME, CONFIG_MAP_NAMESPACE)
+        cm.data["WVA_OPTIMIZER_MODE"] = "limited"
+        cm.data["WVA_SATURATION_POLICY"] = policy
+        cluster.update(cm)
+        acc_cm = cluster.get(ConfigMap, ACCELERATOR_COSTS_CM, CONFIG_MAP_NAMESPACE)
+        for name in list(acc_cm.data):
+            entry = _json.loads(acc_cm.data[name])
+            entry["capacity"] = str(capacity if name == "MI355X" else 0)
+            acc_cm.data[name] = _json.dumps(entry)
+        cluster.update(acc_cm)
+
+    def test_capacity_caps_allocation(self, cluster, prom, registry):
+        # high load wants many replicas; capacity of 1 MI355X unit with
+        # policy None -> no feasible allocation for the variant
+        self._enable_limited(cluster, policy="None", capacity=1)
+        make_deployment(cluster, replicas=1)
+        make_va(cluster)
+        set_load_metrics(prom, "default/llama-8b", "default", arrival_rps=100.0, out_tokens=200.0)
+        VariantAutoscalingReconciler(cluster, prom).reconcile()
+        va = get_va(cluster)
+        # the desired allocation cannot exceed the capacity; with None
+        # policy the variant simply gets nothing (solution empty ->
+        # OptimizationReady False on this cycle)
+        assert va.status.desired_optimized_alloc.num_replicas == 0
+
+    def test_priority_exhaustive_grants_capacity(self, cluster, prom, registry):
+        self._enable_limited(cluster, policy="PriorityExhaustive", capacity=2)
+        make_deployment(cluster, replicas=1)
+        make_va(cluster)
+        set_load_metrics(prom, "default/llama-8b", "default", arrival_rps=100.0, out_tokens=200.0)
+        VariantAutoscalingReconciler(cluster, prom).reconcile()
+        va = get_va(cluster)
+        assert va.status.desired_optimized_alloc.num_replicas == 2  # all capacity

@@ -101,9 +101,9 @@ class VariantAutoscalingReconciler:
         self.analyzer_device = analyzer_device
 
     # ------------------------------------------------------------- config IO
-    def _read_optimization_config(self) -> str:
+    def _read_optimization_config(self) -> Dict[str, str]:
         cm = get_configmap_with_backoff(self.client, CONFIG_MAP_NAME, CONFIG_MAP_NAMESPACE)
-        return cm.data.get("GLOBAL_OPT_INTERVAL", "")
+        return cm.data
 
     def _read_accelerator_config(self) -> Dict[str, Dict[str, str]]:
         cm = get_configmap_with_backoff(self.client, ACCELERATOR_COSTS_CM, CONFIG_MAP_NAMESPACE)
@@ -124,7 +124,8 @@ class VariantAutoscalingReconciler:
 
     # -------------------------------------------------------------- reconcile
     def reconcile(self) -> ReconcileResult:
-        interval = self._read_optimization_config()
+        optimization_cm = self._read_optimization_config()
+        interval = optimization_cm.get("GLOBAL_OPT_INTERVAL", "")
         requeue = DEFAULT_REQUEUE_SECONDS
         if interval:
             requeue = parse_go_duration(interval)
@@ -144,7 +145,7 @@ class VariantAutoscalingReconciler:
             log.info("No active VariantAutoscalings found, skipping optimization")
             return ReconcileResult(None)
 
-        system_data = create_system_data(accelerator_cm, service_class_cm)
+        system_data = create_system_data(accelerator_cm, service_class_cm, optimization_cm)
         update_list, va_map, responses = self._prepare_variant_autoscalings(
             active, accelerator_cm, service_class_cm, system_data
         )

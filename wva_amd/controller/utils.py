@@ -2,7 +2,8 @@
 
 Parity with /root/reference/internal/utils/utils.go: exponential-backoff
 presets and retry wrappers (non-retryable NotFound/Invalid/Forbidden),
-CreateSystemData (Unlimited:true hardwired, utils.go:170-173),
+CreateSystemData (unlimited by default; unlike the reference, limited
+mode is reachable via the controller ConfigMap — see create_system_data),
 AddModelAcceleratorProfileToSystemData (string alpha/beta/gamma/delta
 parsing), AddServerInfoToSystemData (KeepAccelerator:true pinning,
 scale-to-zero via WVA_SCALE_TO_ZERO, maxBatchSize from the matching
@@ -23,6 +24,7 @@ import yaml
 
 from ..api import v1alpha1
 from ..config import (
+    AcceleratorCount,
     AcceleratorSpec,
     AllocationData,
     AllocationSolution,
@@ -128,11 +130,20 @@ def full_name(name: str, namespace: str) -> str:
 
 
 def create_system_data(
-    accelerator_cm: Dict[str, Dict[str, str]], service_class_cm: Dict[str, str]
+    accelerator_cm: Dict[str, Dict[str, str]],
+    service_class_cm: Dict[str, str],
+    optimizer_cm: Optional[Dict[str, str]] = None,
 ) -> SystemData:
     """Adapter from ConfigMap payloads to the optimizer's SystemSpec.
-    Unlimited mode hardwired; capacity unused (utils.go:108-182)."""
+
+    The reference hardwires Unlimited:true and leaves the greedy
+    capacity-constrained solver dormant (utils.go:170-173).  Here limited
+    mode is reachable: set ``WVA_OPTIMIZER_MODE: limited`` in the
+    controller ConfigMap (plus optional ``WVA_SATURATION_POLICY`` and
+    ``WVA_DELAYED_BEST_EFFORT``) and give accelerator entries a
+    ``capacity`` field (units of that type available in the pool)."""
     sd = SystemData(spec=SystemSpec())
+    optimizer_cm = optimizer_cm or {}
 
     for key, val in accelerator_cm.items():
         try:
@@ -153,6 +164,13 @@ def create_system_data(
         except ValueError:
             pass
         sd.spec.accelerators.spec.append(spec)
+        if "capacity" in val:
+            try:
+                sd.spec.capacity.count.append(
+                    AcceleratorCount(type=spec.type, count=int(val["capacity"]))
+                )
+            except ValueError:
+                log.warn("failed to parse accelerator capacity, ignoring", name=key)
 
     for key, val in service_class_cm.items():
         sc = parse_service_class_yaml(key, val)
@@ -169,7 +187,12 @@ def create_system_data(
             )
         )
 
-    sd.spec.optimizer.spec = OptimizerSpec(unlimited=True)
+    mode = optimizer_cm.get("WVA_OPTIMIZER_MODE", "unlimited").lower()
+    sd.spec.optimizer.spec = OptimizerSpec(
+        unlimited=mode != "limited",
+        delayed_best_effort=optimizer_cm.get("WVA_DELAYED_BEST_EFFORT", "").lower() == "true",
+        saturation_policy=optimizer_cm.get("WVA_SATURATION_POLICY", ""),
+    )
     return sd
 
 
