@@ -99,10 +99,12 @@ def server_spec(
 
 
 def make_spec(servers=None, unlimited=True, capacity=None, saturation_policy="", delayed_best_effort=False):
+    import copy
+
     servers = servers if servers is not None else [server_spec("s1:default")]
     capacity = capacity or []
     return SystemSpec(
-        accelerators=AcceleratorData(spec=[MI355X, MI300X, L40S]),
+        accelerators=AcceleratorData(spec=copy.deepcopy([MI355X, MI300X, L40S])),
         models=ModelData(
             perf_data=[
                 perf("llama-8b", "MI355X", alpha=4.0, beta=0.03, gamma=8.0, delta=0.03),
