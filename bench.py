@@ -229,12 +229,14 @@ def main() -> None:
     barrier_sync()
     elapsed = time.perf_counter() - t0
 
-    # max over ranks
+    # max over ranks (collectives need device tensors under RCCL)
     if dist is not None:
-        t = torch.tensor([elapsed], dtype=torch.float64)
+        coll_device = "cuda" if has_gpu else "cpu"
+        t = torch.tensor([elapsed], dtype=torch.float64, device=coll_device)
         slo = torch.tensor(
             [sum(s["slo_met"] for s in stats), sum(s["total"] for s in stats)],
             dtype=torch.float64,
+            device=coll_device,
         )
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         dist.all_reduce(slo, op=dist.ReduceOp.SUM)
