@@ -33,13 +33,19 @@
 
 namespace wva {
 
-#define WVA_THREADS 256
-#define WVA_WAVES (WVA_THREADS / 64)
+// Two workgroup geometries are instantiated:
+//   - 256 threads (4 wave64): strided sweeps + LDS cross-wave combine;
+//   - 64 threads (1 wave64): barrier-free — the wave executes in lockstep,
+//     reductions are pure __shfl_down chains and the combined values are
+//     broadcast from lane 0 with __shfl (no LDS round-trip).
+// The launcher picks by the WVA_GPU_THREADS env var (default 256).
 
 // Reduction scratch layout (doubles, after cum[max_k] in dynamic LDS):
-//   red[0..WVA_WAVES*5-1]  per-wave partials (S, Ni, Snum, Ninum, eK)
+//   red[0..WAVES*5-1]  per-wave partials (S, Ni, Snum, Ninum, eK)
 
+template <int THREADS>
 struct WgEval {
+  static constexpr int WAVES = THREADS / 64;
   const Parms &p;
   const double *cum;  // LDS, K entries
   double *red;        // LDS scratch
@@ -49,6 +55,8 @@ struct WgEval {
     const int tid = threadIdx.x;
     const int wave = tid >> 6;
     const int lane = tid & 63;
+    (void)wave;
+    (void)lane;
     const double loglam = log(lam);
 
     // max of logp via the concavity closed form, and the significant
@@ -60,14 +68,16 @@ struct WgEval {
     if (m < 0.0) m = 0.0;
     int n_lo, n_hi;
     state_window(cum, loglam, K, n_star, m, &n_lo, &n_hi);
-    __syncthreads();  // red may still be read from a previous eval
+    if constexpr (WAVES > 1) {
+      __syncthreads();  // red may still be read from a previous eval
+    }
 
     // single windowed pass: normalization and moment sums (eK is the
     // state-K boundary term, reduced like the sums — only its owner
     // thread contributes a non-zero partial)
     const int num = p.max_batch;
     double S = 0.0, Ni = 0.0, Snum = 0.0, Ninum = 0.0, eK = 0.0;
-    for (int n = n_lo + tid; n <= n_hi; n += WVA_THREADS) {
+    for (int n = n_lo + tid; n <= n_hi; n += THREADS) {
       double e = exp(log_p(cum, loglam, n) - m);
       S += e;
       Ni += (double)n * e;
@@ -84,21 +94,30 @@ struct WgEval {
       Ninum += __shfl_down(Ninum, off, 64);
       eK += __shfl_down(eK, off, 64);
     }
-    if (lane == 0) {
-      red[wave * 5 + 0] = S;
-      red[wave * 5 + 1] = Ni;
-      red[wave * 5 + 2] = Snum;
-      red[wave * 5 + 3] = Ninum;
-      red[wave * 5 + 4] = eK;
-    }
-    __syncthreads();
-    S = Ni = Snum = Ninum = eK = 0.0;
-    for (int w = 0; w < WVA_WAVES; ++w) {
-      S += red[w * 5 + 0];
-      Ni += red[w * 5 + 1];
-      Snum += red[w * 5 + 2];
-      Ninum += red[w * 5 + 3];
-      eK += red[w * 5 + 4];
+    if constexpr (WAVES == 1) {
+      // single wave: broadcast lane 0's totals to every lane
+      S = __shfl(S, 0, 64);
+      Ni = __shfl(Ni, 0, 64);
+      Snum = __shfl(Snum, 0, 64);
+      Ninum = __shfl(Ninum, 0, 64);
+      eK = __shfl(eK, 0, 64);
+    } else {
+      if (lane == 0) {
+        red[wave * 5 + 0] = S;
+        red[wave * 5 + 1] = Ni;
+        red[wave * 5 + 2] = Snum;
+        red[wave * 5 + 3] = Ninum;
+        red[wave * 5 + 4] = eK;
+      }
+      __syncthreads();
+      S = Ni = Snum = Ninum = eK = 0.0;
+      for (int w = 0; w < WAVES; ++w) {
+        S += red[w * 5 + 0];
+        Ni += red[w * 5 + 1];
+        Snum += red[w * 5 + 2];
+        Ninum += red[w * 5 + 3];
+        eK += red[w * 5 + 4];
+      }
     }
 
     Stats st;
@@ -169,9 +188,9 @@ __device__ int wg_binary_search(double x_min, double x_max, double y_target, F e
   return 0;
 }
 
-extern "C" __global__ void __launch_bounds__(WVA_THREADS)
-    wva_solve_kernel(const double *__restrict__ prob, double *__restrict__ out,
-                     int n_problems, int max_k) {
+template <int THREADS>
+__device__ void solve_body(const double *__restrict__ prob, double *__restrict__ out,
+                           int n_problems, int max_k) {
   const int pid = blockIdx.x;
   if (pid >= n_problems) return;
   const double *pr = prob + (size_t)pid * PROBLEM_FIELDS;
@@ -193,11 +212,11 @@ extern "C" __global__ void __launch_bounds__(WVA_THREADS)
 
   extern __shared__ double smem[];
   double *cum = smem;             // this problem's K entries
-  double *totals = smem + max_k;  // 256 chunk totals
-  double *red = totals + WVA_THREADS;
+  double *totals = smem + max_k;  // THREADS chunk totals
+  double *red = totals + THREADS;
 
   // chunked parallel inclusive scan of log_mu over K states
-  const int chunk = (K + WVA_THREADS - 1) / WVA_THREADS;
+  const int chunk = (K + THREADS - 1) / THREADS;
   const int lo = tid * chunk;
   const int hi = min(lo + chunk, K);
   double acc = 0.0;
@@ -215,7 +234,7 @@ extern "C" __global__ void __launch_bounds__(WVA_THREADS)
   const double lam_min = serv_rate(p, 1) * kEpsilon;  // req/ms
   const double lam_max = serv_rate(p, p.max_batch) * (1.0 - kEpsilon);
 
-  WgEval ev{p, cum, red, K};
+  WgEval<THREADS> ev{p, cum, red, K};
 
   if (tid < RESULT_FIELDS && pid < n_problems) res[tid] = 0.0;
 
@@ -260,11 +279,34 @@ extern "C" __global__ void __launch_bounds__(WVA_THREADS)
   }
 }
 
+extern "C" __global__ void __launch_bounds__(256) wva_solve_kernel_256(
+    const double *__restrict__ prob, double *__restrict__ out, int n_problems,
+    int max_k) {
+  solve_body<256>(prob, out, n_problems, max_k);
+}
+
+extern "C" __global__ void __launch_bounds__(64) wva_solve_kernel_64(
+    const double *__restrict__ prob, double *__restrict__ out, int n_problems,
+    int max_k) {
+  solve_body<64>(prob, out, n_problems, max_k);
+}
+
 }  // namespace wva
+
+#include <cstdlib>
+#include <cstring>
 
 extern "C" void wva_launch_solve(const double *prob, double *out, int n_problems,
                                  int max_k, void *stream) {
-  const size_t smem = (size_t)(max_k + WVA_THREADS + 32) * sizeof(double);
-  hipLaunchKernelGGL(wva::wva_solve_kernel, dim3(n_problems), dim3(WVA_THREADS),
-                     smem, (hipStream_t)stream, prob, out, n_problems, max_k);
+  const char *env = std::getenv("WVA_GPU_THREADS");
+  const bool one_wave = env != nullptr && std::strcmp(env, "64") == 0;
+  const int threads = one_wave ? 64 : 256;
+  const size_t smem = (size_t)(max_k + threads + 32) * sizeof(double);
+  if (one_wave) {
+    hipLaunchKernelGGL(wva::wva_solve_kernel_64, dim3(n_problems), dim3(64), smem,
+                       (hipStream_t)stream, prob, out, n_problems, max_k);
+  } else {
+    hipLaunchKernelGGL(wva::wva_solve_kernel_256, dim3(n_problems), dim3(256), smem,
+                       (hipStream_t)stream, prob, out, n_problems, max_k);
+  }
 }
