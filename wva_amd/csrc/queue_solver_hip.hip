@@ -298,8 +298,10 @@ extern "C" __global__ void __launch_bounds__(64) wva_solve_kernel_64(
 
 extern "C" void wva_launch_solve(const double *prob, double *out, int n_problems,
                                  int max_k, void *stream) {
+  // single-wave is the default: measured 20-31% faster than the 4-wave
+  // geometry at fleet batch sizes (profiles/r01_queue_solver.md)
   const char *env = std::getenv("WVA_GPU_THREADS");
-  const bool one_wave = env != nullptr && std::strcmp(env, "64") == 0;
+  const bool one_wave = env == nullptr || std::strcmp(env, "256") != 0;
   const int threads = one_wave ? 64 : 256;
   const size_t smem = (size_t)(max_k + threads + 32) * sizeof(double);
   if (one_wave) {
