@@ -96,6 +96,14 @@ class TestReconcileHappyPath:
         assert registry.get_sample_value("inferno_desired_ratio", labels) == desired / 2.0
         # the solver latency histogram observed this cycle
         assert registry.get_sample_value("wva_solver_duration_seconds_count") == 1.0
+        # per-phase cycle timing observed
+        for phase in ("config", "prepare", "analyze", "optimize", "apply"):
+            assert (
+                registry.get_sample_value(
+                    "wva_cycle_phase_duration_seconds_count", {"phase": phase}
+                )
+                == 1.0
+            ), phase
 
     def test_high_load_scales_out(self, cluster, prom, registry):
         make_deployment(cluster, replicas=1)

@@ -33,6 +33,9 @@ INFERNO_DESIRED_RATIO = "inferno_desired_ratio"
 # only; pkg/solver/optimizer.go:30-34)
 WVA_SOLVER_DURATION_SECONDS = "wva_solver_duration_seconds"
 
+# Per-phase reconcile-cycle timing (config/prepare/analyze/optimize/apply)
+WVA_CYCLE_PHASE_DURATION_SECONDS = "wva_cycle_phase_duration_seconds"
+
 # Label names
 LABEL_MODEL_NAME = "model_name"
 LABEL_NAMESPACE = "namespace"

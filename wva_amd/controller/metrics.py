@@ -25,6 +25,7 @@ _desired_replicas: Optional[Gauge] = None
 _current_replicas: Optional[Gauge] = None
 _desired_ratio: Optional[Gauge] = None
 _solver_duration: Optional[Histogram] = None
+_cycle_phase_duration: Optional[Histogram] = None
 
 
 def init_metrics(registry: CollectorRegistry) -> None:
@@ -68,18 +69,35 @@ def init_metrics(registry: CollectorRegistry) -> None:
             1e-2, 2.5e-2, 5e-2, 0.1, 0.25, 0.5, 1.0, 2.5,
         ),
     )
+    global _cycle_phase_duration
+    _cycle_phase_duration = Histogram(
+        constants.WVA_CYCLE_PHASE_DURATION_SECONDS,
+        "Wall-clock duration of each reconcile-cycle phase",
+        ["phase"],  # config / prepare / analyze / optimize / apply
+        registry=registry,
+        buckets=(
+            1e-4, 5e-4, 1e-3, 2.5e-3, 5e-3, 1e-2, 2.5e-2, 5e-2,
+            0.1, 0.25, 0.5, 1.0, 2.5, 5.0, 10.0,
+        ),
+    )
 
 
 def reset_metrics() -> None:
     """Drop metric handles (tests use fresh registries)."""
-    global _replica_scaling_total, _desired_replicas, _current_replicas, _desired_ratio, _solver_duration
+    global _replica_scaling_total, _desired_replicas, _current_replicas, _desired_ratio
+    global _solver_duration, _cycle_phase_duration
     _replica_scaling_total = _desired_replicas = _current_replicas = _desired_ratio = None
-    _solver_duration = None
+    _solver_duration = _cycle_phase_duration = None
 
 
 def observe_solver_duration(seconds: float) -> None:
     if _solver_duration is not None:
         _solver_duration.observe(seconds)
+
+
+def observe_cycle_phase(phase: str, seconds: float) -> None:
+    if _cycle_phase_duration is not None:
+        _cycle_phase_duration.labels(phase).observe(seconds)
 
 
 class MetricsEmitter:

@@ -33,6 +33,7 @@ from ..core import System
 from ..kube import ConfigMap, KubeClient, NotFoundError
 from ..solver import Manager, Optimizer
 from . import collector
+from . import metrics as ctrl_metrics
 from .actuator import Actuator
 from .engine import OptimizationError, VariantAutoscalingsEngine
 from .interfaces import ModelAnalyzeResponse, PrometheusConfig
@@ -124,6 +125,14 @@ class VariantAutoscalingReconciler:
 
     # -------------------------------------------------------------- reconcile
     def reconcile(self) -> ReconcileResult:
+        phase_t0 = time.perf_counter()
+
+        def mark_phase(name: str) -> None:
+            nonlocal phase_t0
+            now = time.perf_counter()
+            ctrl_metrics.observe_cycle_phase(name, now - phase_t0)
+            phase_t0 = now
+
         optimization_cm = self._read_optimization_config()
         interval = optimization_cm.get("GLOBAL_OPT_INTERVAL", "")
         requeue = DEFAULT_REQUEUE_SECONDS
@@ -145,10 +154,12 @@ class VariantAutoscalingReconciler:
             log.info("No active VariantAutoscalings found, skipping optimization")
             return ReconcileResult(None)
 
+        mark_phase("config")
         system_data = create_system_data(accelerator_cm, service_class_cm, optimization_cm)
         update_list, va_map, responses = self._prepare_variant_autoscalings(
             active, accelerator_cm, service_class_cm, system_data
         )
+        mark_phase("prepare")
 
         system = System()
         optimizer_spec = system.set_from_spec(system_data.spec)
@@ -171,6 +182,7 @@ class VariantAutoscalingReconciler:
                 continue
             responses[name] = response
 
+        mark_phase("analyze")
         engine = VariantAutoscalingsEngine(manager, system)
         try:
             optimized = engine.optimize(
@@ -196,7 +208,9 @@ class VariantAutoscalingReconciler:
                     )
             return ReconcileResult(requeue)
 
+        mark_phase("optimize")
         self._apply_optimized_allocations(update_list, optimized)
+        mark_phase("apply")
         return ReconcileResult(requeue)
 
     # ---------------------------------------------------------------- prepare
