@@ -77,7 +77,8 @@ class TestConfigMaps:
 
 class TestSampleVA:
     def test_round_trips_through_api_types(self):
-        doc = load("samples/mi355x-variantautoscaling.yaml")
+        docs = load("samples/mi355x-variantautoscaling.yaml")
+        doc = docs[0] if isinstance(docs, list) else docs
         va = v1alpha1.VariantAutoscaling.model_validate(doc)
         assert va.spec.model_id == "default/llama-3.1-8b"
         assert va.metadata.labels["inference.optimization/acceleratorName"] == "MI355X"
@@ -88,6 +89,15 @@ class TestSampleVA:
         out = va.to_dict()
         assert out["spec"]["modelID"] == "default/llama-3.1-8b"
         assert "maxBatchSize" in out["spec"]["modelProfile"]["accelerators"][0]
+
+    def test_70b_sample_single_gpu(self):
+        docs = load("samples/mi355x-variantautoscaling.yaml")
+        va70 = v1alpha1.VariantAutoscaling.model_validate(docs[1])
+        assert va70.spec.model_id == "default/llama-3.3-70b"
+        profile = va70.spec.model_profile.accelerators[0]
+        # 70B on ONE MI355X: 288 GB holds the whole model, accCount 1
+        assert profile.acc == "MI355X" and profile.acc_count == 1
+        assert float(profile.perf_parms.decode_parms["alpha"]) > 0
 
 
 class TestIntegrations:
