@@ -7,8 +7,6 @@ concurrency, M/M/1/K probability sums, a Little's-law property check, and
 binary-search boundary classification/precision.
 """
 
-import math
-
 import numpy as np
 import pytest
 

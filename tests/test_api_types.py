@@ -1,8 +1,6 @@
 """API type tests (mirrors api/v1alpha1/variantautoscaling_types_test.go:
 validation behavior, condition transition semantics, serialization)."""
 
-import datetime
-
 import pytest
 from pydantic import ValidationError
 

@@ -7,7 +7,6 @@ with threads (the closest Python analog of running `go test -race`)."""
 import concurrent.futures as cf
 
 import numpy as np
-import pytest
 
 from wva_amd.core import System
 from wva_amd.solver import Manager, Optimizer

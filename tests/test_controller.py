@@ -7,7 +7,7 @@ from prometheus_client import CollectorRegistry
 
 from wva_amd.api import v1alpha1
 from wva_amd.controller import collector, metrics as ctrl_metrics
-from wva_amd.controller.promclient import MockPromAPI, PromQueryError
+from wva_amd.controller.promclient import MockPromAPI
 from wva_amd.controller.reconciler import (
     CONFIG_MAP_NAMESPACE,
     SERVICE_CLASSES_CM,

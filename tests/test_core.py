@@ -5,7 +5,6 @@ import math
 
 import pytest
 
-from wva_amd.config import ServerLoadSpec
 from wva_amd.core import Accelerator, Allocation, System, create_allocation
 from fixtures import MI355X, make_system, server_spec
 

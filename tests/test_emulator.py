@@ -5,7 +5,7 @@ import asyncio
 
 import pytest
 
-from vllm_emulator.engine import Clock, EmulatedVLLM, EmulatorSettings, RequestElement
+from vllm_emulator.engine import EmulatedVLLM, EmulatorSettings, RequestElement
 from vllm_emulator.metrics import EmulatorMetrics
 
 

@@ -5,7 +5,6 @@ from file), and the mandatory-HTTPS validation paths.
 The e2e analog of the reference's TLS handshake checks
 (test/e2e/e2e_test.go:563-628) without a cluster."""
 
-import json
 import subprocess
 import threading
 import time

@@ -7,7 +7,6 @@ import pytest
 
 from prometheus_client import CollectorRegistry
 
-from wva_amd.config import OptimizerSpec
 from wva_amd.controller import metrics as ctrl_metrics
 from wva_amd.core import System
 from wva_amd.solver import Manager, Optimizer, Solver

@@ -17,7 +17,7 @@ import argparse
 import json
 import subprocess
 import time
-from typing import Dict, List, Optional
+from typing import List, Optional
 
 from prometheus_client import CollectorRegistry, Gauge, start_http_server
 

@@ -13,8 +13,6 @@ from pathlib import Path
 sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
 sys.path.insert(0, str(Path(__file__).resolve().parent.parent / "tests"))
 
-import numpy as np
-
 from wva_amd.ops import native_available, solve_problems
 from wva_amd.ops.batched import R_FEASIBLE, _solve_problems_python
 from test_ops import random_problems  # reuse the generator

@@ -13,7 +13,7 @@ import asyncio
 import random
 import time
 from dataclasses import dataclass
-from typing import List, Optional, Sequence, Tuple
+from typing import List, Optional, Sequence
 
 
 @dataclass

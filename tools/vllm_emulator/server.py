@@ -18,7 +18,7 @@ from typing import List, Optional
 from fastapi import FastAPI, Response
 from pydantic import BaseModel
 
-from .engine import Clock, EmulatedVLLM, EmulatorSettings, RequestElement
+from .engine import EmulatedVLLM, EmulatorSettings, RequestElement
 from .metrics import EmulatorMetrics
 
 

@@ -12,7 +12,7 @@ inside the queue model; latencies are milliseconds everywhere.
 from __future__ import annotations
 
 import json
-from dataclasses import asdict, dataclass, field, fields, is_dataclass
+from dataclasses import dataclass, field, fields, is_dataclass
 from typing import Any, Dict, List, get_args, get_origin, get_type_hints
 
 

@@ -24,7 +24,7 @@ from ..api import v1alpha1
 from ..kube import Deployment
 from . import constants
 from .logger import log
-from .promclient import PromAPI, PromQueryError, Sample
+from .promclient import PromAPI, PromQueryError
 
 STALENESS_LIMIT_SECONDS = 5 * 60
 

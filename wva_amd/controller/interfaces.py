@@ -8,7 +8,7 @@ response shapes, the service-class ConfigMap YAML shapes (``slo-tpot`` /
 from __future__ import annotations
 
 from dataclasses import dataclass, field
-from typing import Dict, List, Optional
+from typing import Dict, List
 
 from ..core import Allocation
 

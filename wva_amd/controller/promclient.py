@@ -15,7 +15,7 @@ import os
 import ssl
 import time
 from dataclasses import dataclass, field
-from typing import Dict, List, Optional, Protocol
+from typing import Dict, List, Protocol
 
 from .interfaces import PrometheusConfig
 

@@ -26,7 +26,7 @@ import json
 import re
 import threading
 import time
-from typing import Dict, Optional, Tuple
+from typing import Dict, Optional
 
 from ..api import v1alpha1
 from ..core import System

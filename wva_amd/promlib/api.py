@@ -3,7 +3,7 @@
 from __future__ import annotations
 
 import time
-from typing import List, Optional
+from typing import List
 
 from ..controller.promclient import PromQueryError, Sample
 from .promql import PromQLError, evaluate
