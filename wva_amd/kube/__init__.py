@@ -15,6 +15,7 @@ ownerReference garbage collection, resourceVersion bumping) behind a small
 from .errors import ConflictError, ForbiddenError, InvalidError, KubeError, NotFoundError
 from .objects import ConfigMap, Deployment, DeploymentSpec, DeploymentStatus
 from .client import InMemoryKubeClient, KubeClient
+from .http_client import HTTPKubeClient
 
 __all__ = [
     "KubeError",
@@ -28,4 +29,5 @@ __all__ = [
     "DeploymentStatus",
     "KubeClient",
     "InMemoryKubeClient",
+    "HTTPKubeClient",
 ]

@@ -1,0 +1,182 @@
+"""HTTP Kubernetes client — the real-cluster implementation of the
+KubeClient protocol.
+
+Speaks the API-server REST conventions directly over httpx (no external
+kubernetes package exists in this environment):
+
+- core/v1 ConfigMaps, apps/v1 Deployments, llmd.ai/v1alpha1
+  VariantAutoscalings (+ the /status subresource);
+- in-cluster configuration from the mounted service account
+  (KUBERNETES_SERVICE_HOST/PORT + token + CA), or explicit base_url/token;
+- error mapping onto the package's retryability classes: 404 NotFound,
+  403 Forbidden, 409 Conflict, 400/422 Invalid.
+
+The controller is agnostic: it only sees the protocol
+(wva_amd/kube/client.py), so the in-memory fake and this client are
+interchangeable (exercised by tests/test_kube_http.py against a stub API
+server).
+"""
+
+from __future__ import annotations
+
+import os
+import ssl
+from typing import List, Optional, Type, TypeVar
+
+from ..api.v1alpha1.types import VariantAutoscaling
+from .errors import ConflictError, ForbiddenError, InvalidError, KubeError, NotFoundError
+from .objects import ConfigMap, Deployment
+
+T = TypeVar("T")
+
+SA_DIR = "/var/run/secrets/kubernetes.io/serviceaccount"
+
+_RESOURCES = {
+    VariantAutoscaling: ("apis/llmd.ai/v1alpha1", "variantautoscalings", True),
+    ConfigMap: ("api/v1", "configmaps", False),
+    Deployment: ("apis/apps/v1", "deployments", True),
+}
+
+
+def in_cluster_config() -> dict:
+    host = os.environ.get("KUBERNETES_SERVICE_HOST")
+    port = os.environ.get("KUBERNETES_SERVICE_PORT", "443")
+    if not host:
+        raise RuntimeError("not running in a cluster (KUBERNETES_SERVICE_HOST unset)")
+    with open(f"{SA_DIR}/token") as f:
+        token = f.read().strip()
+    return {
+        "base_url": f"https://{host}:{port}",
+        "token": token,
+        "ca_cert_path": f"{SA_DIR}/ca.crt",
+    }
+
+
+class HTTPKubeClient:
+    def __init__(
+        self,
+        base_url: Optional[str] = None,
+        token: str = "",
+        ca_cert_path: str = "",
+        verify: bool = True,
+    ) -> None:
+        import httpx
+
+        if base_url is None:
+            config = in_cluster_config()
+            base_url, token, ca_cert_path = (
+                config["base_url"],
+                config["token"],
+                config["ca_cert_path"],
+            )
+        headers = {"Authorization": f"Bearer {token}"} if token else {}
+        if base_url.startswith("https"):
+            if verify and ca_cert_path:
+                ctx = ssl.SSLContext(ssl.PROTOCOL_TLS_CLIENT)
+                ctx.minimum_version = ssl.TLSVersion.TLSv1_2
+                ctx.load_verify_locations(cafile=ca_cert_path)
+                verify_arg = ctx
+            elif verify:
+                verify_arg = True
+            else:
+                verify_arg = False
+        else:
+            verify_arg = False
+        self._client = httpx.Client(
+            base_url=base_url, headers=headers, verify=verify_arg, timeout=10.0
+        )
+
+    # ------------------------------------------------------------------ paths
+    @staticmethod
+    def _resource(cls_or_obj):
+        cls = cls_or_obj if isinstance(cls_or_obj, type) else type(cls_or_obj)
+        try:
+            return _RESOURCES[cls]
+        except KeyError:
+            raise TypeError(f"unregistered kind {cls.__name__}") from None
+
+    def _path(self, cls_or_obj, namespace: str, name: str = "", subresource: str = "") -> str:
+        prefix, plural, _ = self._resource(cls_or_obj)
+        path = f"/{prefix}/namespaces/{namespace}/{plural}"
+        if name:
+            path += f"/{name}"
+        if subresource:
+            path += f"/{subresource}"
+        return path
+
+    @staticmethod
+    def _raise_for(resp, what: str) -> None:
+        if resp.status_code < 400:
+            return
+        msg = f"{what}: {resp.status_code} {resp.text[:300]}"
+        if resp.status_code == 404:
+            raise NotFoundError(msg)
+        if resp.status_code == 403:
+            raise ForbiddenError(msg)
+        if resp.status_code == 409:
+            raise ConflictError(msg)
+        if resp.status_code in (400, 422):
+            raise InvalidError(msg)
+        raise KubeError(msg)
+
+    @staticmethod
+    def _dump(obj) -> dict:
+        d = obj.model_dump(by_alias=True, exclude_none=True, mode="json")
+        rv = d.get("metadata", {}).get("resourceVersion")
+        if rv is not None:
+            # the API server requires resourceVersion as a string
+            d["metadata"]["resourceVersion"] = str(rv)
+        return d
+
+    # ------------------------------------------------------------------- CRUD
+    def get(self, cls: Type[T], name: str, namespace: str) -> T:
+        resp = self._client.get(self._path(cls, namespace, name))
+        self._raise_for(resp, f"get {cls.__name__} {namespace}/{name}")
+        return cls.model_validate(resp.json())
+
+    def list(self, cls: Type[T], namespace: Optional[str] = None) -> List[T]:
+        prefix, plural, namespaced = self._resource(cls)
+        if namespace is None:
+            path = f"/{prefix}/{plural}"
+        else:
+            path = f"/{prefix}/namespaces/{namespace}/{plural}"
+        resp = self._client.get(path)
+        self._raise_for(resp, f"list {cls.__name__}")
+        return [cls.model_validate(item) for item in resp.json().get("items", [])]
+
+    def create(self, obj: T) -> T:
+        resp = self._client.post(
+            self._path(obj, obj.metadata.namespace), json=self._dump(obj)
+        )
+        self._raise_for(resp, f"create {type(obj).__name__}")
+        return type(obj).model_validate(resp.json())
+
+    def update(self, obj: T) -> T:
+        resp = self._client.put(
+            self._path(obj, obj.metadata.namespace, obj.metadata.name),
+            json=self._dump(obj),
+        )
+        self._raise_for(resp, f"update {type(obj).__name__}")
+        return type(obj).model_validate(resp.json())
+
+    def patch_metadata(self, obj: T) -> T:
+        patch = {"metadata": self._dump(obj)["metadata"]}
+        resp = self._client.patch(
+            self._path(obj, obj.metadata.namespace, obj.metadata.name),
+            json=patch,
+            headers={"Content-Type": "application/merge-patch+json"},
+        )
+        self._raise_for(resp, f"patch {type(obj).__name__}")
+        return type(obj).model_validate(resp.json())
+
+    def update_status(self, obj: T) -> T:
+        resp = self._client.put(
+            self._path(obj, obj.metadata.namespace, obj.metadata.name, "status"),
+            json=self._dump(obj),
+        )
+        self._raise_for(resp, f"update status {type(obj).__name__}")
+        return type(obj).model_validate(resp.json())
+
+    def delete(self, cls: Type[T], name: str, namespace: str) -> None:
+        resp = self._client.delete(self._path(cls, namespace, name))
+        self._raise_for(resp, f"delete {cls.__name__} {namespace}/{name}")
