@@ -49,7 +49,12 @@ def parse_args(argv=None):
     ap.add_argument("--metrics-cert-name", default="tls.crt")
     ap.add_argument("--metrics-cert-key", default="tls.key")
     ap.add_argument("--enable-http2", action="store_true", default=False)
-    ap.add_argument("--kube-backend", choices=["memory"], default="memory")
+    ap.add_argument(
+        "--kube-backend",
+        choices=["memory", "in-cluster"],
+        default="memory",
+        help="'in-cluster' uses the mounted service account against the real API server",
+    )
     ap.add_argument("--max-cycles", type=int, default=None, help="exit after N reconcile cycles")
     return ap.parse_args(argv)
 
@@ -114,7 +119,12 @@ def main(argv=None) -> int:
     registry = CollectorRegistry()
     ctrl_metrics.init_metrics(registry)
 
-    client = InMemoryKubeClient()  # --kube-backend memory
+    if args.kube_backend == "in-cluster":
+        from .kube import HTTPKubeClient
+
+        client = HTTPKubeClient()  # service-account config
+    else:
+        client = InMemoryKubeClient()
 
     ready = {"ok": False}
     server = serve_http(args, registry, lambda: ready["ok"])

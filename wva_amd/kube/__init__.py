@@ -8,8 +8,9 @@ ownerReference garbage collection, resourceVersion bumping) behind a small
 
 - :class:`InMemoryKubeClient` — the envtest analog used by the component
   and e2e test tiers (SURVEY.md §4.2-4.3);
-- a real HTTP client can be slotted in later without touching the
-  controller (the controller only sees the protocol).
+- :class:`HTTPKubeClient` — the real-cluster client speaking the API
+  server's REST conventions (in-cluster service-account config or
+  explicit endpoint); the controller only sees the protocol.
 """
 
 from .errors import ConflictError, ForbiddenError, InvalidError, KubeError, NotFoundError
