@@ -88,6 +88,10 @@ def build_system_spec(rank: int, step: int, n_variants: int) -> SystemSpec:
     """Synthetic cluster state for this rank at this step (loads vary per
     step so every cycle is a fresh solve)."""
     rng = np.random.default_rng(10_000 * (rank + 1) + step)
+    replicas = rng.integers(1, 8, n_variants)
+    rates = rng.uniform(30.0, 36000.0, n_variants)  # req/min
+    in_tokens = rng.integers(64, 2048, n_variants)
+    out_tokens = rng.integers(32, 1024, n_variants)
     servers = []
     for i in range(n_variants):
         model = MODELS[i % len(MODELS)]
@@ -102,11 +106,11 @@ def build_system_spec(rank: int, step: int, n_variants: int) -> SystemSpec:
                 max_batch_size=256,
                 current_alloc=AllocationData(
                     accelerator="MI355X",
-                    num_replicas=int(rng.integers(1, 8)),
+                    num_replicas=int(replicas[i]),
                     load=ServerLoadSpec(
-                        arrival_rate=float(rng.uniform(30.0, 36000.0)),  # req/min
-                        avg_in_tokens=int(rng.integers(64, 2048)),
-                        avg_out_tokens=int(rng.integers(32, 1024)),
+                        arrival_rate=float(rates[i]),
+                        avg_in_tokens=int(in_tokens[i]),
+                        avg_out_tokens=int(out_tokens[i]),
                     ),
                 ),
             )
