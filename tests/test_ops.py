@@ -146,3 +146,22 @@ class TestGPU:
         out = solve_problems(problems, device="cuda")
         ref = solve_problems(problems, device="cpu")
         assert_results_close(out, ref, rtol=1e-4)
+
+
+@pytest.mark.gpu
+def test_fuzz_parity_gpu():
+    """Wide-range fuzz (subset of tools/fuzz_parity.py) as a regression
+    gate: zero feasibility flips, zero replica differences."""
+    import sys
+    from pathlib import Path
+
+    sys.path.insert(0, str(Path(__file__).resolve().parent.parent / "tools"))
+    from fuzz_parity import draw
+
+    problems = draw(4000, seed=99)
+    gpu = solve_problems(problems, device="cuda")
+    cpu = solve_problems(problems, device="cpu")
+    assert np.isfinite(gpu).all() and np.isfinite(cpu).all()
+    assert (gpu[:, R_FEASIBLE] == cpu[:, R_FEASIBLE]).all()
+    both = gpu[:, R_FEASIBLE] == 1
+    assert (gpu[both, R_REPLICAS] == cpu[both, R_REPLICAS]).all()
