@@ -1,0 +1,100 @@
+"""Actuator suite (mirrors internal/actuator tests: replica-source
+fallbacks, metric emission, ratio scenarios)."""
+
+import pytest
+from prometheus_client import CollectorRegistry
+
+from wva_amd.api import v1alpha1
+from wva_amd.api.v1alpha1.types import ObjectMeta
+from wva_amd.controller import metrics as ctrl_metrics
+from wva_amd.controller.actuator import Actuator
+from wva_amd.kube import Deployment, DeploymentSpec, DeploymentStatus, InMemoryKubeClient
+
+
+@pytest.fixture()
+def registry():
+    reg = CollectorRegistry()
+    ctrl_metrics.init_metrics(reg)
+    yield reg
+    ctrl_metrics.reset_metrics()
+
+
+def make_va(name="v", namespace="ns", desired=4, desired_acc="MI355X", current=2):
+    va = v1alpha1.VariantAutoscaling(metadata=ObjectMeta(name=name, namespace=namespace))
+    va.status.desired_optimized_alloc = v1alpha1.OptimizedAlloc(
+        accelerator=desired_acc, numReplicas=desired
+    )
+    va.status.current_alloc.num_replicas = current
+    return va
+
+
+def gauge(registry, name, va, acc="MI355X"):
+    return registry.get_sample_value(
+        name, {"variant_name": va.name, "namespace": va.namespace, "accelerator_type": acc}
+    )
+
+
+class TestReplicaSource:
+    def test_prefers_deployment_status(self, registry):
+        client = InMemoryKubeClient()
+        client.create(
+            Deployment(
+                metadata=ObjectMeta(name="v", namespace="ns"),
+                spec=DeploymentSpec(replicas=7),
+                status=DeploymentStatus(replicas=3),
+            )
+        )
+        va = make_va()
+        Actuator(client).emit_metrics(va)
+        assert gauge(registry, "inferno_current_replicas", va) == 3.0
+        assert gauge(registry, "inferno_desired_replicas", va) == 4.0
+
+    def test_falls_back_to_va_status_when_deployment_missing(self, registry):
+        client = InMemoryKubeClient()
+        va = make_va(current=5)
+        Actuator(client).emit_metrics(va)  # no Deployment exists
+        assert gauge(registry, "inferno_current_replicas", va) == 5.0
+
+    def test_ratio_normal(self, registry):
+        client = InMemoryKubeClient()
+        client.create(
+            Deployment(
+                metadata=ObjectMeta(name="v", namespace="ns"),
+                status=DeploymentStatus(replicas=2),
+            )
+        )
+        va = make_va(desired=6)
+        Actuator(client).emit_metrics(va)
+        assert gauge(registry, "inferno_desired_ratio", va) == 3.0
+
+    def test_ratio_zero_to_n_encoding(self, registry):
+        client = InMemoryKubeClient()
+        client.create(
+            Deployment(
+                metadata=ObjectMeta(name="v", namespace="ns"),
+                status=DeploymentStatus(replicas=0),
+            )
+        )
+        va = make_va(desired=5)
+        Actuator(client).emit_metrics(va)
+        # 0 -> N is encoded as ratio = N (metrics.go:118-124 parity)
+        assert gauge(registry, "inferno_desired_ratio", va) == 5.0
+
+    def test_emitter_failure_does_not_raise(self):
+        # metrics not initialized: the emitter raises internally, the
+        # actuator must swallow it (reconciliation never breaks on metrics)
+        ctrl_metrics.reset_metrics()
+        client = InMemoryKubeClient()
+        va = make_va()
+        Actuator(client).emit_metrics(va)  # no exception
+
+    def test_scaling_counter(self, registry):
+        va = make_va()
+        ctrl_metrics.MetricsEmitter().emit_replica_scaling_metrics(va, "up", "slo")
+        assert (
+            registry.get_sample_value(
+                "inferno_replica_scaling_total",
+                {"variant_name": "v", "namespace": "ns", "direction": "up", "reason": "slo"},
+            )
+            == 1.0
+        )
