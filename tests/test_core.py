@@ -199,3 +199,17 @@ class TestScaleAndReallocate:
         system, _ = make_system()
         alloc, acc = reallocate(system, "missing")
         assert alloc is None and acc == ""
+
+
+class TestDefaultServiceClass:
+    def test_empty_class_name_defaults_to_free(self):
+        from wva_amd.config import DEFAULT_SERVICE_CLASS_NAME
+
+        system, _ = make_system(servers=[server_spec("s:ns", class_name="")])
+        server = system.server("s:ns")
+        assert server.service_class_name == DEFAULT_SERVICE_CLASS_NAME == "Free"
+        # no "Free" class registered -> no feasible allocation (parity with
+        # GetServiceClass nil -> CreateAllocation nil)
+        assert create_allocation(system, "s:ns", "MI355X") is None
+        # priority falls back to the default lowest
+        assert server.priority(system) == 100
