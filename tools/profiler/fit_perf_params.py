@@ -32,15 +32,51 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 
+class FP8Linear(nn.Module):
+    """Linear layer computing in OCP e4m3fn via torch._scaled_mm (the
+    hipBLASLt fp8 path on gfx950): weights quantized once column-wise,
+    activations quantized row-wise per forward — the standard quantized-
+    serving recipe, so the measured curves reflect fp8 deployment."""
+
+    def __init__(self, in_features: int, out_features: int) -> None:
+        super().__init__()
+        w = torch.randn(out_features, in_features) * (in_features**-0.5)
+        amax = w.abs().amax(dim=1, keepdim=True).clamp(min=1e-6)
+        scale = amax / 448.0  # e4m3fn max normal
+        self.register_buffer("weight_fp8", (w / scale).to(torch.float8_e4m3fn))
+        self.register_buffer("weight_scale", scale.T.float())  # [1, out]
+
+    def forward(self, x):
+        shape = x.shape
+        x2 = x.reshape(-1, shape[-1])
+        amax = x2.abs().amax(dim=1, keepdim=True).clamp(min=1e-6).float()
+        scale_a = amax / 448.0
+        x8 = (x2 / scale_a.to(x2.dtype)).to(torch.float8_e4m3fn)
+        out = torch._scaled_mm(
+            x8,
+            self.weight_fp8.t(),
+            scale_a=scale_a,
+            scale_b=self.weight_scale,
+            out_dtype=torch.bfloat16,
+        )
+        return out.reshape(*shape[:-1], -1)
+
+
 class DecoderLayer(nn.Module):
-    def __init__(self, hidden: int, heads: int, ffn_mult: int = 4) -> None:
+    def __init__(self, hidden: int, heads: int, ffn_mult: int = 4, linear_cls=nn.Linear) -> None:
         super().__init__()
         self.heads = heads
         self.head_dim = hidden // heads
-        self.qkv = nn.Linear(hidden, 3 * hidden, bias=False)
-        self.o = nn.Linear(hidden, hidden, bias=False)
-        self.up = nn.Linear(hidden, ffn_mult * hidden, bias=False)
-        self.down = nn.Linear(ffn_mult * hidden, hidden, bias=False)
+        if linear_cls is nn.Linear:
+            self.qkv = nn.Linear(hidden, 3 * hidden, bias=False)
+            self.o = nn.Linear(hidden, hidden, bias=False)
+            self.up = nn.Linear(hidden, ffn_mult * hidden, bias=False)
+            self.down = nn.Linear(ffn_mult * hidden, hidden, bias=False)
+        else:
+            self.qkv = linear_cls(hidden, 3 * hidden)
+            self.o = linear_cls(hidden, hidden)
+            self.up = linear_cls(hidden, ffn_mult * hidden)
+            self.down = linear_cls(ffn_mult * hidden, hidden)
         self.norm1 = nn.LayerNorm(hidden)
         self.norm2 = nn.LayerNorm(hidden)
 
@@ -63,10 +99,12 @@ class DecoderLayer(nn.Module):
 
 
 class TinyDecoder(nn.Module):
-    def __init__(self, layers: int, hidden: int, heads: int, vocab: int = 32000) -> None:
+    def __init__(self, layers: int, hidden: int, heads: int, vocab: int = 32000, linear_cls=nn.Linear) -> None:
         super().__init__()
         self.embed = nn.Embedding(vocab, hidden)
-        self.layers = nn.ModuleList(DecoderLayer(hidden, heads) for _ in range(layers))
+        self.layers = nn.ModuleList(
+            DecoderLayer(hidden, heads, linear_cls=linear_cls) for _ in range(layers)
+        )
         self.head = nn.Linear(hidden, vocab, bias=False)
 
     def prefill(self, tokens):
@@ -168,11 +206,20 @@ def fit(
     warmup: int = 3,
     device: str = None,
     dtype=torch.bfloat16,
+    fp8: bool = False,
 ) -> FitResult:
     device = device or ("cuda" if torch.cuda.is_available() else "cpu")
     if device == "cpu":
         dtype = torch.float32
-    model = TinyDecoder(layers, hidden, heads).to(device=device, dtype=dtype).eval()
+    linear_cls = FP8Linear if fp8 else nn.Linear
+    model = TinyDecoder(layers, hidden, heads, linear_cls=linear_cls)
+    model = model.to(device=device, dtype=dtype).eval()
+    if fp8:
+        # .to(dtype) converted the fp8 buffers; restore their dtypes
+        for module in model.modules():
+            if isinstance(module, FP8Linear):
+                module.weight_fp8 = module.weight_fp8.to(torch.float8_e4m3fn)
+                module.weight_scale = module.weight_scale.float()
     return fit_with_model(model, device, list(batches), seq_len, decode_iters, warmup)
 
 
@@ -190,6 +237,7 @@ def main() -> None:
     ap.add_argument("--decode-iters", type=int, default=40)
     ap.add_argument("--warmup", type=int, default=5)
     ap.add_argument("--acc", default="MI355X")
+    ap.add_argument("--dtype", choices=["bf16", "fp8"], default="bf16")
     ap.add_argument("--out", default="")
     args = ap.parse_args()
 
@@ -201,10 +249,11 @@ def main() -> None:
         seq_len=args.seq_len,
         decode_iters=args.decode_iters,
         warmup=args.warmup,
+        fp8=args.dtype == "fp8",
     )
     payload = {
         "acc": args.acc,
-        "model": f"tiny-decoder-L{args.layers}-H{args.hidden}",
+        "model": f"tiny-decoder-L{args.layers}-H{args.hidden}-{args.dtype}",
         "perfParms": {
             "decodeParms": {"alpha": f"{result.alpha:.4f}", "beta": f"{result.beta:.6f}"},
             "prefillParms": {"gamma": f"{result.gamma:.4f}", "delta": f"{result.delta:.8f}"},
