@@ -227,6 +227,11 @@ class OptimizerSpec(SpecBase):
     unlimited: bool = jfield("unlimited", False)
     delayed_best_effort: bool = jfield("delayedBestEffort", False)
     saturation_policy: str = jfield("saturationPolicy", "")
+    # objective extension: "" / "cost" (default) or "cost+energy" — the
+    # latter adds predicted power draw (accelerator power curve at the
+    # allocation's utilization) priced at energyCostPerKWh cents/kWh
+    objective: str = jfield("objective", "")
+    energy_cost_per_kwh: float = jfield("energyCostPerKWh", 0.0)
 
 
 @dataclass

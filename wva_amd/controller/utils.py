@@ -188,10 +188,16 @@ def create_system_data(
         )
 
     mode = optimizer_cm.get("WVA_OPTIMIZER_MODE", "unlimited").lower()
+    try:
+        energy_price = float(optimizer_cm.get("WVA_ENERGY_COST_PER_KWH", "0"))
+    except ValueError:
+        energy_price = 0.0
     sd.spec.optimizer.spec = OptimizerSpec(
         unlimited=mode != "limited",
         delayed_best_effort=optimizer_cm.get("WVA_DELAYED_BEST_EFFORT", "").lower() == "true",
         saturation_policy=optimizer_cm.get("WVA_SATURATION_POLICY", ""),
+        objective=optimizer_cm.get("WVA_OBJECTIVE", ""),
+        energy_cost_per_kwh=energy_price,
     )
     return sd
 

@@ -217,6 +217,30 @@ def create_allocation(system: "System", server_name: str, acc_name: str) -> Opti
     return alloc
 
 
+def energy_value_term(system: "System", server, alloc: Allocation) -> float:
+    """Energy surcharge for the ``cost+energy`` objective (an MI355X-native
+    extension: the reference computes the accelerator power curve but never
+    uses it — accelerator.go:35-41 / SURVEY.md §2a).
+
+    Predicted draw = power(rho) per card x cards per replica x replicas,
+    priced at energyCostPerKWh (cents/kWh) -> cents/hr, the same unit as
+    the accelerator cost, so it composes with cost- and penalty-based
+    values.  Returns 0 unless the objective is enabled.
+    """
+    spec = system.optimizer_spec
+    if spec is None or spec.objective != "cost+energy" or spec.energy_cost_per_kwh <= 0:
+        return 0.0
+    if alloc.num_replicas == 0 or not alloc.accelerator:
+        return 0.0
+    acc = system.accelerator(alloc.accelerator)
+    model = system.model(server.model_name)
+    if acc is None or model is None:
+        return 0.0
+    cards = model.get_num_instances(alloc.accelerator) * acc.multiplicity * alloc.num_replicas
+    watts = acc.power(alloc.rho) * cards
+    return watts / 1000.0 * spec.energy_cost_per_kwh  # cents per hour
+
+
 def scale_allocation(
     system: "System", alloc: Allocation, server_name: str
 ) -> Tuple[Optional[Allocation], int]:

@@ -15,7 +15,7 @@ from ..config import (
     ServerLoadSpec,
     ServerSpec,
 )
-from .allocation import Allocation, create_allocation
+from .allocation import Allocation, create_allocation, energy_value_term
 
 if TYPE_CHECKING:  # pragma: no cover
     from .accelerator import Accelerator
@@ -38,7 +38,8 @@ class Server:
 
     def calculate(self, system: "System", accelerators: Dict[str, "Accelerator"]) -> None:
         """Enumerate candidate allocations; value = transition penalty from
-        the current allocation (server.go:55-67)."""
+        the current allocation (server.go:55-67), plus the optional
+        energy term of the cost+energy objective."""
         candidates = self.get_candidate_accelerators(accelerators)
         self.all_allocations = {}
         for g in candidates.values():
@@ -46,6 +47,7 @@ class Server:
             if alloc is not None:
                 if self.cur_allocation is not None:
                     alloc.set_value(self.cur_allocation.transition_penalty(alloc))
+                alloc.set_value(alloc.value + energy_value_term(system, self, alloc))
                 self.all_allocations[g.name] = alloc
 
     def get_candidate_accelerators(

@@ -49,6 +49,7 @@ class System:
         self.capacity: Dict[str, int] = {}
         self.allocation_by_type: Dict[str, AllocationByType] = {}
         self.allocation_solution: Optional[AllocationSolution] = None
+        self.optimizer_spec: Optional[OptimizerSpec] = None
 
     # -- spec loading -------------------------------------------------------
     def set_from_spec(self, spec: SystemSpec) -> OptimizerSpec:
@@ -62,6 +63,7 @@ class System:
             self.add_server(srv)
         for cnt in spec.capacity.count:
             self.set_capacity(cnt)
+        self.optimizer_spec = spec.optimizer.spec
         return spec.optimizer.spec
 
     def add_accelerator(self, spec: AcceleratorSpec) -> None:

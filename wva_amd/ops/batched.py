@@ -29,7 +29,7 @@ from ..analyzer import (
 )
 from ..config import MAX_QUEUE_TO_BATCH_RATIO
 from ..core import Allocation, System
-from ..core.allocation import _zero_load_allocation
+from ..core.allocation import _zero_load_allocation, energy_value_term
 
 # problem columns
 P_ALPHA, P_BETA, P_GAMMA, P_DELTA = 0, 1, 2, 3
@@ -224,6 +224,7 @@ class BatchedAllocationSolver:
             alloc.set_value(alloc.cost)
             if server.cur_allocation is not None:
                 alloc.set_value(server.cur_allocation.transition_penalty(alloc))
+            alloc.set_value(alloc.value + energy_value_term(system, server, alloc))
             server.all_allocations[acc_name] = alloc
 
         for server_name, accs in zero_load.items():
@@ -231,4 +232,5 @@ class BatchedAllocationSolver:
             for acc_name, alloc in accs.items():
                 if server.cur_allocation is not None:
                     alloc.set_value(server.cur_allocation.transition_penalty(alloc))
+                alloc.set_value(alloc.value + energy_value_term(system, server, alloc))
                 server.all_allocations[acc_name] = alloc
