@@ -112,14 +112,26 @@ struct Stats {
 // concave and its maximum sits where the increment first turns <= 0:
 // an O(log N) search over the monotone service-rate curve instead of an
 // O(K) max sweep.  Returns the state index n* in [0, K].
+// serv_rate(b) < lam  <=>  b < lam * T(b)  (service time T(b) > 0):
+// a multiply instead of the fp64 divide inside the mode bisection
+WVA_HD bool rate_below(const Parms &p, int b, double lam) {
+  double t = prefill_time(p, (double)b) + p.num_decode * decode_time(p, (double)b);
+  return (double)b < lam * t;
+}
+
 WVA_HD int log_mode_state(const Parms &p, int K, double lam) {
-  if (lam >= serv_rate(p, p.max_batch)) return K;  // all increments > 0
-  if (lam <= serv_rate(p, 1)) return 0;
+  // lam >= mu(N)  <=>  N <= lam * T(N): all increments > 0, mode at K
+  double tN = prefill_time(p, (double)p.max_batch) +
+              p.num_decode * decode_time(p, (double)p.max_batch);
+  if ((double)p.max_batch <= lam * tN) return K;
+  // lam <= mu(1)  <=>  lam * T(1) <= 1: mode at 0
+  double t1 = prefill_time(p, 1.0) + p.num_decode * decode_time(p, 1.0);
+  if (lam * t1 <= 1.0) return 0;
   // smallest b in [1, N] with serv_rate(b) >= lam; mode = b - 1
   int lo = 1, hi = p.max_batch;
   while (lo < hi) {
     int mid = (lo + hi) / 2;
-    if (serv_rate(p, mid) < lam) {
+    if (rate_below(p, mid, lam)) {
       lo = mid + 1;
     } else {
       hi = mid;
