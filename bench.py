@@ -192,6 +192,7 @@ def main() -> None:
     ap.add_argument("--steps", type=int, default=20)
     ap.add_argument("--warmup", type=int, default=5)
     ap.add_argument("--variants-per-gpu", type=int, default=VARIANTS_PER_GPU)
+    ap.add_argument("--backend", default="", help="override torch.distributed backend (default: nccl on GPU, gloo on CPU)")
     args = ap.parse_args()
 
     world_size = int(os.environ.get("WORLD_SIZE", "1"))
@@ -210,7 +211,8 @@ def main() -> None:
     if world_size > 1:
         import torch.distributed as dist_mod
 
-        dist_mod.init_process_group(backend="nccl" if has_gpu else "gloo")
+        backend = args.backend or ("nccl" if has_gpu else "gloo")
+        dist_mod.init_process_group(backend=backend)
         dist = dist_mod
 
     def barrier_sync():
@@ -235,7 +237,7 @@ def main() -> None:
 
     # max over ranks (collectives need device tensors under RCCL)
     if dist is not None:
-        coll_device = "cuda" if has_gpu else "cpu"
+        coll_device = "cuda" if (has_gpu and dist.get_backend() == "nccl") else "cpu"
         t = torch.tensor([elapsed], dtype=torch.float64, device=coll_device)
         slo = torch.tensor(
             [sum(s["slo_met"] for s in stats), sum(s["total"] for s in stats)],
