@@ -51,7 +51,7 @@ class TestBench:
                 "--nnodes=1", "--nproc-per-node", "2",
                 "--master-addr", "127.0.0.1", "--master-port", "29517",
                 "bench.py", "--gpus", "2", "--steps", "2", "--warmup", "1",
-                "--variants-per-gpu", "8",
+                "--variants-per-gpu", "8", "--backend", "gloo",
             ],
             cwd=ROOT,
             capture_output=True,
