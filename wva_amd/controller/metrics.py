@@ -29,7 +29,11 @@ _cycle_phase_duration: Optional[Histogram] = None
 
 
 def init_metrics(registry: CollectorRegistry) -> None:
-    """Register all custom metrics with the provided registry."""
+    """Register all custom metrics with the provided registry and hook the
+    solver's duration observer."""
+    from ..solver.optimizer import register_solve_observer
+
+    register_solve_observer(observe_solver_duration)
     global _replica_scaling_total, _desired_replicas, _current_replicas, _desired_ratio, _solver_duration
     _replica_scaling_total = Counter(
         constants.INFERNO_REPLICA_SCALING_TOTAL,
