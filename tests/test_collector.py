@@ -1,7 +1,6 @@
 """Collector suite (mirrors internal/collector tests: query/status tests
 with MockPromAPI, unit conversions, status string formats)."""
 
-import pytest
 
 from wva_amd.api import v1alpha1
 from wva_amd.api.v1alpha1.types import ObjectMeta

@@ -146,7 +146,6 @@ class TestMainStartup:
         import threading
         import time
 
-        import httpx
         import pytest as _pytest
 
         uvicorn = _pytest.importorskip("uvicorn")

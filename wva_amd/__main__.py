@@ -21,7 +21,6 @@ from __future__ import annotations
 import argparse
 import fcntl
 import os
-import ssl
 import sys
 import threading
 
