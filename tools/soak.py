@@ -26,6 +26,7 @@ def main() -> None:
     ap = argparse.ArgumentParser()
     ap.add_argument("--stages", type=float, nargs="+", default=[2.0, 4.0, 6.0, 4.0, 2.0, 0.0])
     ap.add_argument("--stage-seconds", type=float, default=8.0)
+    ap.add_argument("--variants", type=int, default=1, help="number of variants/emulators")
     ap.add_argument("--out", default="")
     args = ap.parse_args()
 
@@ -54,20 +55,17 @@ def main() -> None:
     ctrl_metrics.init_metrics(registry)
 
     device = "cuda" if torch.cuda.is_available() else "cpu"
-    settings = EmulatorSettings(
-        model="default/llama-8b",
-        decode_alpha=12.0,
-        decode_beta=6.0,
-        prefill_gamma=4.0,
-        prefill_delta=0.01,
-        avg_generated_len=25,
-        tokens_distribution="deterministic",
-        max_batch_size=16,
-        realtime=True,
-    )
     cluster = make_cluster(opt_interval="1s")
-    make_deployment(cluster, replicas=1)
-    make_va(cluster, max_batch=16, alpha="12.0", beta="6.0", gamma="4.0", delta="0.01")
+    # variant 0 runs the staircase; additional variants carry steady load
+    names, models = [], []
+    for v in range(args.variants):
+        name = "vllm-llama" if v == 0 else f"vllm-extra-{v}"
+        model = "default/llama-8b" if v == 0 else f"default/llama-70b"
+        names.append(name)
+        models.append(model)
+        make_deployment(cluster, name=name, replicas=1)
+        make_va(cluster, name=name, model_id=model, max_batch=16,
+                alpha="12.0", beta="6.0", gamma="4.0", delta="0.01")
 
     store = TimeSeriesStore()
     scraper = Scraper(store)
@@ -78,22 +76,51 @@ def main() -> None:
         analyzer_device=device if device == "cuda" else None,
     )
 
+    import contextlib
+
     trajectory = []
-    with EmulatorProcess(settings) as emu:
-        scraper.add_target(f"{emu.base_url}/metrics", extra_labels={"namespace": "default"})
+    with contextlib.ExitStack() as stack:
+        emus = []
+        for v, model in enumerate(models):
+            settings = EmulatorSettings(
+                model=model,
+                decode_alpha=12.0,
+                decode_beta=6.0,
+                prefill_gamma=4.0,
+                prefill_delta=0.01,
+                avg_generated_len=25,
+                tokens_distribution="deterministic",
+                max_batch_size=16,
+                realtime=True,
+            )
+            emu = stack.enter_context(EmulatorProcess(settings))
+            scraper.add_target(f"{emu.base_url}/metrics", extra_labels={"namespace": "default"})
+            emus.append(emu)
         scraper.start(interval=0.5)
         try:
             for rate in args.stages:
-                if rate > 0:
-                    drive_load(emu.base_url, rate_rps=rate, duration_s=args.stage_seconds)
-                else:
+                import threading
+
+                threads = []
+                for v, emu in enumerate(emus):
+                    r = rate if v == 0 else 2.0  # extras: steady 2 rps
+                    if r > 0:
+                        t = threading.Thread(
+                            target=drive_load, args=(emu.base_url, r, args.stage_seconds)
+                        )
+                        t.start()
+                        threads.append(t)
+                if not threads:
                     time.sleep(max(args.stage_seconds, 10.0))
+                for t in threads:
+                    t.join()
                 t0 = time.perf_counter()
                 rec.reconcile()
                 cycle_ms = (time.perf_counter() - t0) * 1000.0
                 va = cluster.get(v1alpha1.VariantAutoscaling, "vllm-llama", "default")
                 desired = va.status.desired_optimized_alloc.num_replicas
-                simulate_hpa(cluster, registry)
+                for name in names:
+                    simulate_hpa(cluster, registry, name=name)
                 entry = {
                     "offered_rps": rate,
                     "measured_arrival_rpm": float(va.status.current_alloc.load.arrival_rate),
@@ -102,6 +129,11 @@ def main() -> None:
                     "metrics_ok": v1alpha1.is_condition_true(va, v1alpha1.TYPE_METRICS_AVAILABLE),
                     "optimized_ok": v1alpha1.is_condition_true(va, v1alpha1.TYPE_OPTIMIZATION_READY),
                 }
+                if args.variants > 1:
+                    entry["extra_desired"] = [
+                        cluster.get(v1alpha1.VariantAutoscaling, n, "default").status.desired_optimized_alloc.num_replicas
+                        for n in names[1:]
+                    ]
                 trajectory.append(entry)
                 print(json.dumps(entry))
         finally:
