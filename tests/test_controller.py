@@ -424,3 +424,23 @@ class TestLimitedMode:
         VariantAutoscalingReconciler(cluster, prom).reconcile()
         va = get_va(cluster)
         assert va.status.desired_optimized_alloc.num_replicas == 2  # all capacity
+
+
+class TestAcceleratorLabelEdge:
+    def test_label_without_matching_profile_skips_gracefully(self, cluster, prom, registry):
+        """VA labeled with an accelerator that has a unit cost but no perf
+        profile row: keep-accelerator pins candidates to the labeled type,
+        no perf data exists for it, so the variant gets no allocation —
+        and the cycle survives (quirk chain through utils.go:296-307 +
+        server.go:70-82)."""
+        make_deployment(cluster)
+        va = make_va(cluster, accelerator="MI355X")
+        # flip the label to a priced-but-unprofiled accelerator
+        va.metadata.labels["inference.optimization/acceleratorName"] = "L40S"
+        cluster.update(va)
+        set_load_metrics(prom, "default/llama-8b", "default", arrival_rps=5.0)
+        VariantAutoscalingReconciler(cluster, prom).reconcile()
+        out = get_va(cluster)
+        assert out.status.desired_optimized_alloc.accelerator == ""
+        # currentAlloc was still collected (metrics fine)
+        assert float(out.status.current_alloc.load.arrival_rate) > 0

@@ -102,3 +102,22 @@ class TestAmdSmiExporter:
         exporter = AmdSmiExporter()
         n = exporter.collect_once()
         assert n >= 1  # at least one MI355X visible
+
+
+class TestDriftCheck:
+    def test_load_profile_and_check(self):
+        from profiler.drift_check import check, load_profile, relative_drift
+
+        configured = load_profile("deploy/samples/mi355x-variantautoscaling.yaml", "MI355X")
+        assert configured["alpha"] == 4.95
+        assert relative_drift(10.0, 12.5) == pytest.approx(0.25)
+        ok = check(configured, dict(configured), tolerance=0.25)
+        assert ok["ok"] and ok["drifted"] == []
+        drifted = check(configured, {**configured, "alpha": configured["alpha"] * 2}, 0.25)
+        assert not drifted["ok"] and drifted["drifted"] == ["alpha"]
+
+    def test_70b_doc_index(self):
+        from profiler.drift_check import load_profile
+
+        p70 = load_profile("deploy/samples/mi355x-variantautoscaling.yaml", "MI355X", doc_index=1)
+        assert p70["alpha"] == 30.06
