@@ -104,11 +104,17 @@ class TestAmdSmiExporter:
         assert n >= 1  # at least one MI355X visible
 
 
+SAMPLE_VA = str(
+    __import__("pathlib").Path(__file__).resolve().parent.parent
+    / "deploy/samples/mi355x-variantautoscaling.yaml"
+)
+
+
 class TestDriftCheck:
     def test_load_profile_and_check(self):
         from profiler.drift_check import check, load_profile, relative_drift
 
-        configured = load_profile("deploy/samples/mi355x-variantautoscaling.yaml", "MI355X")
+        configured = load_profile(SAMPLE_VA, "MI355X")
         assert configured["alpha"] == 4.95
         assert relative_drift(10.0, 12.5) == pytest.approx(0.25)
         ok = check(configured, dict(configured), tolerance=0.25)
@@ -119,5 +125,5 @@ class TestDriftCheck:
     def test_70b_doc_index(self):
         from profiler.drift_check import load_profile
 
-        p70 = load_profile("deploy/samples/mi355x-variantautoscaling.yaml", "MI355X", doc_index=1)
+        p70 = load_profile(SAMPLE_VA, "MI355X", doc_index=1)
         assert p70["alpha"] == 30.06

@@ -88,7 +88,7 @@ def main() -> None:
         "alpha": result.alpha,
         "beta": result.beta,
         "gamma": result.gamma,
-        "delta": result.delta * args.seq_len / args.seq_len,  # per-token slope
+        "delta": result.delta,  # per-token slope
     }
     report = check(configured, measured, args.tolerance)
     print(json.dumps(report, indent=2))
