@@ -444,3 +444,28 @@ class TestAcceleratorLabelEdge:
         assert out.status.desired_optimized_alloc.accelerator == ""
         # currentAlloc was still collected (metrics fine)
         assert float(out.status.current_alloc.load.arrival_rate) > 0
+
+
+class TestGpuTelemetry:
+    def test_amd_smi_signals_collected_per_cycle(self, cluster, prom, registry):
+        make_deployment(cluster)
+        make_va(cluster)
+        set_load_metrics(prom, "default/llama-8b", "default", arrival_rps=2.0)
+        prom.set_result('avg(amd_smi_gpu_gfx_activity{namespace="default"})', 73.5)
+        prom.set_result('sum(amd_smi_gpu_vram_used_bytes{namespace="default"})', 2.0e11)
+        prom.set_result('sum(amd_smi_gpu_power_watts{namespace="default"})', 980.0)
+        rec = VariantAutoscalingReconciler(cluster, prom)
+        rec.reconcile()
+        telemetry = rec.last_gpu_telemetry["default"]
+        assert telemetry.utilization_pct == 73.5
+        assert telemetry.power_watts == 980.0
+
+    def test_absent_exporter_is_harmless(self, cluster, prom, registry):
+        make_deployment(cluster)
+        make_va(cluster)
+        set_load_metrics(prom, "default/llama-8b", "default", arrival_rps=2.0)
+        rec = VariantAutoscalingReconciler(cluster, prom)
+        rec.reconcile()  # MockPromAPI returns zeros for unknown series
+        assert rec.last_gpu_telemetry == {}
+        va = get_va(cluster)
+        assert va.status.desired_optimized_alloc.num_replicas >= 1

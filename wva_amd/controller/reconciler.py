@@ -100,6 +100,7 @@ class VariantAutoscalingReconciler:
         self.prom_api = prom_api
         self.batched_analyzer = batched_analyzer
         self.analyzer_device = analyzer_device
+        self.last_gpu_telemetry = {}
 
     # ------------------------------------------------------------- config IO
     def _read_optimization_config(self) -> Dict[str, str]:
@@ -159,6 +160,24 @@ class VariantAutoscalingReconciler:
         update_list, va_map, responses = self._prepare_variant_autoscalings(
             active, accelerator_cm, service_class_cm, system_data
         )
+
+        # auxiliary MI355X telemetry (amd-smi exporter) per namespace —
+        # best-effort observability alongside the vLLM signals; absent
+        # exporters cost one query and change nothing
+        self.last_gpu_telemetry = {}
+        for namespace in sorted({va.namespace for va in active}):
+            telemetry = collector.collect_gpu_telemetry(self.prom_api, namespace)
+            if telemetry is not None and (
+                telemetry.utilization_pct or telemetry.vram_used_bytes or telemetry.power_watts
+            ):
+                self.last_gpu_telemetry[namespace] = telemetry
+                log.debug(
+                    "GPU telemetry",
+                    namespace=namespace,
+                    gfx_pct=telemetry.utilization_pct,
+                    vram_gib=telemetry.vram_used_bytes / 2**30,
+                    power_w=telemetry.power_watts,
+                )
         mark_phase("prepare")
 
         system = System()
