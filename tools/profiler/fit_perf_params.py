@@ -98,13 +98,74 @@ class DecoderLayer(nn.Module):
         return x, (k, v)
 
 
+class MoELayer(nn.Module):
+    """Mixtral-style sparse FFN: top-2 of E experts per token, tokens
+    grouped per expert (the serving-style dispatch, so measured decode
+    curves reflect real MoE batching behavior on CDNA4)."""
+
+    def __init__(self, hidden: int, experts: int = 8, top_k: int = 2, ffn_mult: int = 4) -> None:
+        super().__init__()
+        self.experts = experts
+        self.top_k = top_k
+        self.gate = nn.Linear(hidden, experts, bias=False)
+        self.w_up = nn.Parameter(torch.randn(experts, hidden, ffn_mult * hidden) * hidden**-0.5)
+        self.w_down = nn.Parameter(torch.randn(experts, ffn_mult * hidden, hidden) * (ffn_mult * hidden) ** -0.5)
+
+    def forward(self, x):
+        B, T, H = x.shape
+        flat = x.reshape(-1, H)
+        logits = self.gate(flat)
+        weights, topk = logits.topk(self.top_k, dim=-1)
+        weights = torch.softmax(weights, dim=-1, dtype=torch.float32).to(x.dtype)
+        out = torch.zeros_like(flat)
+        for e in range(self.experts):
+            mask = topk == e  # [N, top_k]
+            token_idx, slot_idx = mask.nonzero(as_tuple=True)
+            if token_idx.numel() == 0:
+                continue
+            tokens = flat[token_idx]
+            h = F.silu(tokens @ self.w_up[e]) @ self.w_down[e]
+            out.index_add_(0, token_idx, h * weights[token_idx, slot_idx, None])
+        return out.reshape(B, T, H)
+
+
+class MoEDecoderLayer(DecoderLayer):
+    def __init__(self, hidden: int, heads: int, experts: int = 8) -> None:
+        super().__init__(hidden, heads)
+        self.moe = MoELayer(hidden, experts=experts)
+
+    def forward(self, x, kv_cache=None):
+        B, T, H = x.shape
+        residual = x
+        x = self.norm1(x)
+        qkv = self.qkv(x).view(B, T, 3, self.heads, self.head_dim)
+        q, k, v = qkv.unbind(2)
+        q, k, v = (t.transpose(1, 2) for t in (q, k, v))
+        if kv_cache is not None:
+            k = torch.cat([kv_cache[0], k], dim=2)
+            v = torch.cat([kv_cache[1], v], dim=2)
+        attn = F.scaled_dot_product_attention(q, k, v, is_causal=kv_cache is None)
+        x = self.o(attn.transpose(1, 2).reshape(B, T, H))
+        x = residual + x
+        x = x + self.moe(self.norm2(x))
+        return x, (k, v)
+
+
 class TinyDecoder(nn.Module):
-    def __init__(self, layers: int, hidden: int, heads: int, vocab: int = 32000, linear_cls=nn.Linear) -> None:
+    def __init__(
+        self, layers: int, hidden: int, heads: int, vocab: int = 32000,
+        linear_cls=nn.Linear, moe_experts: int = 0,
+    ) -> None:
         super().__init__()
         self.embed = nn.Embedding(vocab, hidden)
-        self.layers = nn.ModuleList(
-            DecoderLayer(hidden, heads, linear_cls=linear_cls) for _ in range(layers)
-        )
+        if moe_experts > 0:
+            self.layers = nn.ModuleList(
+                MoEDecoderLayer(hidden, heads, experts=moe_experts) for _ in range(layers)
+            )
+        else:
+            self.layers = nn.ModuleList(
+                DecoderLayer(hidden, heads, linear_cls=linear_cls) for _ in range(layers)
+            )
         self.head = nn.Linear(hidden, vocab, bias=False)
 
     def prefill(self, tokens):
@@ -207,12 +268,13 @@ def fit(
     device: str = None,
     dtype=torch.bfloat16,
     fp8: bool = False,
+    moe_experts: int = 0,
 ) -> FitResult:
     device = device or ("cuda" if torch.cuda.is_available() else "cpu")
     if device == "cpu":
         dtype = torch.float32
     linear_cls = FP8Linear if fp8 else nn.Linear
-    model = TinyDecoder(layers, hidden, heads, linear_cls=linear_cls)
+    model = TinyDecoder(layers, hidden, heads, linear_cls=linear_cls, moe_experts=moe_experts)
     model = model.to(device=device, dtype=dtype).eval()
     if fp8:
         # .to(dtype) converted the fp8 buffers; restore their dtypes
@@ -238,6 +300,8 @@ def main() -> None:
     ap.add_argument("--warmup", type=int, default=5)
     ap.add_argument("--acc", default="MI355X")
     ap.add_argument("--dtype", choices=["bf16", "fp8"], default="bf16")
+    ap.add_argument("--moe-experts", type=int, default=0,
+                    help="top-2 MoE with this many experts per layer (0 = dense)")
     ap.add_argument("--out", default="")
     args = ap.parse_args()
 
@@ -250,10 +314,12 @@ def main() -> None:
         decode_iters=args.decode_iters,
         warmup=args.warmup,
         fp8=args.dtype == "fp8",
+        moe_experts=args.moe_experts,
     )
     payload = {
         "acc": args.acc,
-        "model": f"tiny-decoder-L{args.layers}-H{args.hidden}-{args.dtype}",
+        "model": f"tiny-decoder-L{args.layers}-H{args.hidden}-{args.dtype}"
+        + (f"-moe{args.moe_experts}x" if args.moe_experts else ""),
         "perfParms": {
             "decodeParms": {"alpha": f"{result.alpha:.4f}", "beta": f"{result.beta:.6f}"},
             "prefillParms": {"gamma": f"{result.gamma:.4f}", "delta": f"{result.delta:.8f}"},
