@@ -429,3 +429,39 @@ class TestEnergyObjective:
         for acc, a in scalar_sys.server("s:ns").all_allocations.items():
             b = batch_sys.server("s:ns").all_allocations[acc]
             assert b.value == pytest.approx(a.value, rel=1e-6)
+
+
+class TestSolveCLI:
+    def test_solve_spec_roundtrip(self, tmp_path):
+        import json
+        import subprocess
+        import sys as _sys
+
+        from fixtures import make_spec
+
+        spec = make_spec(
+            servers=[server_spec("a:ns", arrival_rate=600.0), server_spec("b:ns", arrival_rate=60.0)]
+        )
+        path = tmp_path / "system.json"
+        path.write_text(json.dumps({"system": spec.to_dict()}))
+        proc = subprocess.run(
+            [_sys.executable, "-m", "wva_amd.solve", str(path)],
+            capture_output=True,
+            text=True,
+            timeout=120,
+            cwd=str(__import__("pathlib").Path(__file__).resolve().parent.parent),
+        )
+        assert proc.returncode == 0, proc.stderr
+        out = json.loads(proc.stdout)
+        assert set(out["allocations"]) == {"a:ns", "b:ns"}
+        assert out["solutionTimeMsec"] >= 0
+        assert out["allocations"]["a:ns"]["numReplicas"] >= 1
+        assert out["unallocated"] == []
+
+    def test_bare_spec_accepted(self):
+        from wva_amd.solve import load_spec, solve_spec
+        from fixtures import make_spec
+
+        spec = make_spec()
+        result = solve_spec(load_spec(spec.to_json()))
+        assert "s1:default" in result["allocations"]
