@@ -127,3 +127,36 @@ class TestDriftCheck:
 
         p70 = load_profile(SAMPLE_VA, "MI355X", doc_index=1)
         assert p70["alpha"] == 30.06
+
+
+class TestMoELayer:
+    def test_topk_weights_and_shapes(self):
+        import torch
+
+        from profiler.fit_perf_params import MoELayer
+
+        torch.manual_seed(0)
+        layer = MoELayer(hidden=32, experts=4, top_k=2).eval()
+        x = torch.randn(2, 3, 32)
+        with torch.no_grad():
+            out = layer(x)
+        assert out.shape == x.shape
+        assert torch.isfinite(out).all()
+
+    def test_identical_experts_match_dense_ffn(self):
+        import torch
+        import torch.nn.functional as F
+
+        from profiler.fit_perf_params import MoELayer
+
+        torch.manual_seed(1)
+        layer = MoELayer(hidden=16, experts=4, top_k=2).eval()
+        with torch.no_grad():
+            # make every expert identical: routing becomes irrelevant and
+            # the MoE must reduce to the single dense FFN
+            layer.w_up.copy_(layer.w_up[0].expand_as(layer.w_up))
+            layer.w_down.copy_(layer.w_down[0].expand_as(layer.w_down))
+            x = torch.randn(1, 5, 16)
+            moe_out = layer(x)
+            dense = F.silu(x @ layer.w_up[0]) @ layer.w_down[0]
+        assert torch.allclose(moe_out, dense, atol=1e-5)
