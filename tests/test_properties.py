@@ -1,0 +1,155 @@
+"""Property-based invariants (hypothesis) for the analytics and solver
+layers — randomized counterparts of the reference's table-driven suites."""
+
+import math
+
+import numpy as np
+import pytest
+from hypothesis import given, settings, strategies as st
+
+from wva_amd.analyzer import (
+    AnalyzerError,
+    Configuration,
+    DecodeParms,
+    PrefillParms,
+    QueueAnalyzer,
+    RequestSize,
+    ServiceParms,
+    TargetPerf,
+)
+from fixtures import make_system, server_spec
+
+parms = st.fixed_dictionaries(
+    {
+        "alpha": st.floats(0.5, 100.0),
+        "beta": st.floats(0.001, 5.0),
+        "gamma": st.floats(0.5, 100.0),
+        "delta": st.floats(1e-5, 0.5),
+        "max_batch": st.integers(1, 128),
+        "in_tokens": st.integers(0, 4096),
+        "out_tokens": st.integers(1, 512),
+    }
+)
+
+
+def analyzer_of(p):
+    return QueueAnalyzer(
+        Configuration(
+            max_batch_size=p["max_batch"],
+            max_queue_size=10 * p["max_batch"],
+            service_parms=ServiceParms(
+                prefill=PrefillParms(p["gamma"], p["delta"]),
+                decode=DecodeParms(p["alpha"], p["beta"]),
+            ),
+        ),
+        RequestSize(p["in_tokens"], p["out_tokens"]),
+    )
+
+
+class TestAnalyzerProperties:
+    @settings(max_examples=30, deadline=None)
+    @given(parms)
+    def test_service_rates_monotone_and_positive(self, p):
+        qa = analyzer_of(p)
+        assert (qa.serv_rate > 0).all()
+        assert (np.diff(qa.serv_rate) >= -1e-15).all()  # b/(c+db) is nondecreasing
+
+    @settings(max_examples=30, deadline=None)
+    @given(parms, st.floats(0.01, 0.99))
+    def test_analysis_invariants(self, p, frac):
+        qa = analyzer_of(p)
+        rate = qa.rate_range.min + frac * (qa.rate_range.max - qa.rate_range.min)
+        m = qa.analyze(rate)
+        # throughput cannot exceed the offered rate; all stats sane
+        assert 0 < m.throughput <= rate * (1 + 1e-9)
+        assert m.avg_wait_time >= 0
+        assert 0 <= m.rho <= 1
+        assert 0 <= m.avg_num_in_serv <= p["max_batch"] + 1e-9
+        # model distribution is a distribution
+        assert qa.model.p.sum() == pytest.approx(1.0, rel=1e-9)
+        # Little's law on the full system
+        assert qa.model.avg_num_in_system == pytest.approx(
+            qa.model.throughput * qa.model.avg_resp_time, rel=1e-8
+        )
+
+    @settings(max_examples=20, deadline=None)
+    @given(parms)
+    def test_occupancy_monotone_in_rate(self, p):
+        # E[N] of a birth-death chain is stochastically increasing in the
+        # arrival rate.  (Expected WAIT is not monotone in general for
+        # state-dependent service — higher load shifts mass to states with
+        # faster per-customer service — which hypothesis duly found.)
+        qa = analyzer_of(p)
+        rates = np.linspace(qa.rate_range.min, qa.rate_range.max, 6)
+        occupancy = []
+        for r in rates:
+            qa.analyze(float(r))
+            occupancy.append(qa.model.avg_num_in_system)
+        assert all(b >= a - 1e-9 for a, b in zip(occupancy, occupancy[1:]))
+
+    @settings(max_examples=20, deadline=None)
+    @given(parms, st.floats(1.05, 20.0))
+    def test_size_meets_itl_target(self, p, slack):
+        qa = analyzer_of(p)
+        # a target strictly inside the achievable ITL band
+        itl_min = qa._eval_itl(qa.rate_range.min / 1000.0)
+        itl_max = qa._eval_itl(qa.rate_range.max / 1000.0)
+        if not (itl_max > itl_min * 1.01):
+            return  # flat band: nothing to invert
+        target = min(itl_min * slack, itl_max * 0.999)
+        if target <= itl_min:
+            return
+        try:
+            _, metrics, achieved = qa.size(TargetPerf(target_itl=target))
+        except AnalyzerError:
+            return
+        assert achieved.target_itl <= target * (1 + 1e-3)
+
+
+class TestGreedyProperties:
+    @settings(max_examples=15, deadline=None)
+    @given(
+        st.lists(st.floats(30.0, 30000.0), min_size=1, max_size=6),
+        st.integers(0, 40),
+        st.integers(0, 40),
+        st.sampled_from(["None", "PriorityExhaustive", "PriorityRoundRobin", "RoundRobin"]),
+    )
+    def test_capacity_never_exceeded(self, rates, cap355, cap300, policy):
+        from wva_amd.solver import Solver
+
+        servers = [
+            server_spec(
+                f"s{i}:ns",
+                class_name="Premium" if i % 2 == 0 else "Freemium",
+                arrival_rate=r,
+            )
+            for i, r in enumerate(rates)
+        ]
+        system, opt = make_system(
+            servers=servers,
+            unlimited=False,
+            capacity=[("AMD-MI355X-288GB", cap355), ("AMD-MI300X-192GB", cap300)],
+            saturation_policy=policy,
+        )
+        system.remove_accelerator("L40S")
+        system.calculate()
+        Solver(opt).solve(system)
+        system.allocate_by_type()
+        for t, entry in system.allocation_by_type.items():
+            assert entry.count <= system.capacity.get(t, 0), (t, policy)
+
+    @settings(max_examples=15, deadline=None)
+    @given(st.lists(st.floats(30.0, 30000.0), min_size=1, max_size=5))
+    def test_unlimited_is_argmin(self, rates):
+        from wva_amd.solver import Solver
+
+        servers = [server_spec(f"s{i}:ns", arrival_rate=r) for i, r in enumerate(rates)]
+        system, opt = make_system(servers=servers, unlimited=True)
+        system.calculate()
+        Solver(opt).solve(system)
+        for server in system.servers.values():
+            if not server.all_allocations:
+                assert server.allocation is None
+                continue
+            best = min(a.value for a in server.all_allocations.values())
+            assert server.allocation.value == best
