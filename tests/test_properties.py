@@ -1,7 +1,6 @@
 """Property-based invariants (hypothesis) for the analytics and solver
 layers — randomized counterparts of the reference's table-driven suites."""
 
-import math
 
 import numpy as np
 import pytest
