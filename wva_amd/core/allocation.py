@@ -45,6 +45,11 @@ if TYPE_CHECKING:  # pragma: no cover
 class Allocation:
     """Details of an accelerator allocated to a server."""
 
+    __slots__ = (
+        "accelerator", "num_replicas", "batch_size", "cost", "value",
+        "itl", "ttft", "rho", "max_arrv_rate_per_replica",
+    )
+
     def __init__(
         self,
         accelerator: str = "",

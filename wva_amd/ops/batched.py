@@ -140,10 +140,36 @@ class BatchedAllocationSolver:
         for g in system.accelerators.values():
             g.calculate()
 
-        rows: List[List[float]] = []
+        # pre-resolve the per-(model, accelerator) constants once — fleets
+        # repeat model/accelerator combinations heavily
+        pair_cache: Dict[Tuple[str, str], Optional[Tuple[float, float, float, float, int, int, float]]] = {}
+
+        def pair_info(model, acc):
+            key = (model.name, acc.name)
+            hit = pair_cache.get(key, False)
+            if hit is not False:
+                return hit
+            perf = model.get_perf_data(acc.name)
+            if perf is None:
+                pair_cache[key] = None
+                return None
+            info = (
+                perf.decode_parms.alpha,
+                perf.decode_parms.beta,
+                perf.prefill_parms.gamma,
+                perf.prefill_parms.delta,
+                perf.max_batch_size,
+                perf.at_tokens,
+                acc.cost * model.get_num_instances(acc.name),
+            )
+            pair_cache[key] = info
+            return info
+
+        rows: List[Tuple[float, ...]] = []
         keys: List[Tuple[str, str, int]] = []  # (server, acc, N)
         zero_load: Dict[str, Dict[str, Allocation]] = {}
         costs: List[float] = []
+        rows_append, keys_append, costs_append = rows.append, keys.append, costs.append
 
         for server in system.servers.values():
             server.all_allocations = {}
@@ -164,73 +190,71 @@ class BatchedAllocationSolver:
             target = svc.model_target(server.model_name)
             if target is None:
                 continue
+            target_ttft, target_itl, target_tps = target.ttft, target.itl, target.tps
+            arrival, in_tok, out_tok = load.arrival_rate, load.avg_in_tokens, load.avg_out_tokens
+            server_name = server.name
+            server_max_batch = server.max_batch_size
+            min_replicas = float(server.min_num_replicas)
             candidates = server.get_candidate_accelerators(system.accelerators)
             for acc in candidates.values():
-                perf = model.get_perf_data(acc.name)
-                if perf is None:
+                info = pair_info(model, acc)
+                if info is None:
                     continue
-                if load.arrival_rate == 0 or load.avg_out_tokens == 0:
+                if arrival == 0 or out_tok == 0:
+                    perf = model.get_perf_data(acc.name)
                     alloc = _zero_load_allocation(server, model, acc, perf)
-                    zero_load.setdefault(server.name, {})[acc.name] = alloc
+                    zero_load.setdefault(server_name, {})[acc.name] = alloc
                     continue
-                K = int(load.avg_out_tokens)
-                if server.max_batch_size > 0:
-                    N = server.max_batch_size
-                else:
-                    N = max(perf.max_batch_size * perf.at_tokens // K, 1)
-                if target.tps == 0:
-                    total_rate = load.arrival_rate / 60.0
-                else:
-                    total_rate = target.tps / float(K)
-                rows.append(
-                    [
-                        perf.decode_parms.alpha,
-                        perf.decode_parms.beta,
-                        perf.prefill_parms.gamma,
-                        perf.prefill_parms.delta,
-                        float(int(load.avg_in_tokens)),
-                        float(K),
-                        float(N),
-                        target.ttft,
-                        target.itl,
-                        target.tps,
-                        total_rate,
-                        float(server.min_num_replicas),
-                    ]
+                alpha, beta, gamma, delta, perf_max_batch, at_tokens, cost_per_rep = info
+                K = int(out_tok)
+                N = server_max_batch if server_max_batch > 0 else max(perf_max_batch * at_tokens // K, 1)
+                total_rate = arrival / 60.0 if target_tps == 0 else target_tps / float(K)
+                rows_append(
+                    (alpha, beta, gamma, delta, float(int(in_tok)), float(K), float(N),
+                     target_ttft, target_itl, target_tps, total_rate, min_replicas)
                 )
-                keys.append((server.name, acc.name, N))
-                costs.append(acc.cost * model.get_num_instances(acc.name))
+                keys_append((server_name, acc.name, N))
+                costs_append(cost_per_rep)
 
         if rows:
-            results = solve_problems(np.asarray(rows, dtype=np.float64), self.device)
+            results = solve_problems(np.array(rows, dtype=np.float64), self.device)
         else:
             results = np.zeros((0, RESULT_FIELDS))
 
-        for (server_name, acc_name, N), res, cost_per_replica in zip(keys, results, costs):
+        opt_spec = system.optimizer_spec
+        energy_enabled = (
+            opt_spec is not None
+            and opt_spec.objective == "cost+energy"
+            and opt_spec.energy_cost_per_kwh > 0
+        )
+        servers = system.servers
+        results_list = results.tolist()  # plain floats beat numpy scalar access
+        for (server_name, acc_name, N), res, cost_per_replica in zip(keys, results_list, costs):
             if res[R_FEASIBLE] != 1.0:
                 continue
-            server = system.server(server_name)
+            server = servers[server_name]
             num_replicas = int(res[R_REPLICAS])
             alloc = Allocation(
                 accelerator=acc_name,
                 num_replicas=num_replicas,
                 batch_size=N,
                 cost=cost_per_replica * num_replicas,
-                itl=float(res[R_ITL]),
-                ttft=float(res[R_TTFT]),
-                rho=float(res[R_RHO]),
-                max_arrv_rate_per_replica=float(res[R_RATE_STAR]) / 1000.0,
+                itl=res[R_ITL],
+                ttft=res[R_TTFT],
+                rho=res[R_RHO],
+                max_arrv_rate_per_replica=res[R_RATE_STAR] / 1000.0,
             )
-            alloc.set_value(alloc.cost)
-            if server.cur_allocation is not None:
-                alloc.set_value(server.cur_allocation.transition_penalty(alloc))
-            alloc.set_value(alloc.value + energy_value_term(system, server, alloc))
+            cur = server.cur_allocation
+            alloc.value = cur.transition_penalty(alloc) if cur is not None else alloc.cost
+            if energy_enabled:
+                alloc.value += energy_value_term(system, server, alloc)
             server.all_allocations[acc_name] = alloc
 
         for server_name, accs in zero_load.items():
-            server = system.server(server_name)
+            server = servers[server_name]
             for acc_name, alloc in accs.items():
                 if server.cur_allocation is not None:
                     alloc.set_value(server.cur_allocation.transition_penalty(alloc))
-                alloc.set_value(alloc.value + energy_value_term(system, server, alloc))
+                if energy_enabled:
+                    alloc.set_value(alloc.value + energy_value_term(system, server, alloc))
                 server.all_allocations[acc_name] = alloc
