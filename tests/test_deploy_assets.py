@@ -240,3 +240,11 @@ class TestMainStartup:
         finally:
             server.should_exit = True
             thread.join(timeout=5.0)
+
+
+def test_webhook_cert_flags_accepted():
+    from wva_amd.__main__ import parse_args
+
+    args = parse_args(["--webhook-cert-path", "/certs"])
+    assert args.webhook_cert_path == "/certs"
+    assert args.webhook_cert_name == "tls.crt"

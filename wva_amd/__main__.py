@@ -47,6 +47,11 @@ def parse_args(argv=None):
     ap.add_argument("--metrics-cert-path", default="")
     ap.add_argument("--metrics-cert-name", default="tls.crt")
     ap.add_argument("--metrics-cert-key", default="tls.key")
+    # webhook certificate flags accepted for CLI parity; no admission
+    # webhook is served (the reference wires them into its webhook server)
+    ap.add_argument("--webhook-cert-path", default="")
+    ap.add_argument("--webhook-cert-name", default="tls.crt")
+    ap.add_argument("--webhook-cert-key", default="tls.key")
     ap.add_argument("--enable-http2", action="store_true", default=False)
     ap.add_argument(
         "--kube-backend",
