@@ -121,7 +121,7 @@ class TestMainEntry:
         from wva_amd.__main__ import parse_args
 
         args = parse_args([])
-        assert args.metrics_bind_address == ":8443"
+        assert args.metrics_bind_address == "0"  # disabled by default (reference parity)
         assert args.health_probe_bind_address == ":8081"
         assert args.leader_elect is False
         assert args.metrics_secure is True

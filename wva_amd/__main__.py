@@ -36,8 +36,9 @@ LEADER_LOCK_ID = "72dd1cf1.llm-d.ai"
 
 def parse_args(argv=None):
     ap = argparse.ArgumentParser(prog="wva-amd-controller")
-    ap.add_argument("--metrics-bind-address", default=":8443",
-                    help="metrics endpoint bind address ('0' disables)")
+    ap.add_argument("--metrics-bind-address", default="0",
+                    help="metrics endpoint bind address ('0' disables; the "
+                         "deploy manifests pass :8443)")
     ap.add_argument("--health-probe-bind-address", default=":8081")
     ap.add_argument("--leader-elect", action="store_true",
                     help="enable leader election for controller manager")
