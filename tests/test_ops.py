@@ -165,3 +165,31 @@ def test_fuzz_parity_gpu():
     assert (gpu[:, R_FEASIBLE] == cpu[:, R_FEASIBLE]).all()
     both = gpu[:, R_FEASIBLE] == 1
     assert (gpu[both, R_REPLICAS] == cpu[both, R_REPLICAS]).all()
+
+
+@pytest.mark.gpu
+def test_concurrent_stream_submissions():
+    """Two host threads driving the kernel on separate HIP streams must
+    produce the same results as serial execution (extension stream-safety)."""
+    import concurrent.futures as cf
+
+    import torch
+
+    from wva_amd.ops import get_native
+
+    native = get_native()
+    problems = random_problems(512, seed=21)
+    t = torch.from_numpy(problems).cuda()
+    serial = native.solve_allocations(t).cpu().numpy()
+
+    def run_on_stream(_):
+        stream = torch.cuda.Stream()
+        with torch.cuda.stream(stream):
+            out = native.solve_allocations(t)
+        stream.synchronize()
+        return out.cpu().numpy()
+
+    with cf.ThreadPoolExecutor(max_workers=4) as pool:
+        outs = list(pool.map(run_on_stream, range(8)))
+    for out in outs:
+        np.testing.assert_array_equal(out, serial)
