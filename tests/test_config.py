@@ -56,3 +56,14 @@ class TestMI355XCatalog:
         assert entry["device"] == "AMD-MI355X-288GB"
         assert float(entry["cost"]) > 0
         assert entry["memSize"] == "288"
+
+
+class TestPackageAlias:
+    def test_full_name_alias(self):
+        import workload_variant_autoscaler_amd as full
+        import workload_variant_autoscaler_amd.analyzer as full_analyzer
+        import wva_amd
+        import wva_amd.analyzer
+
+        assert full is wva_amd
+        assert full_analyzer is wva_amd.analyzer
