@@ -138,6 +138,7 @@ def main(argv=None) -> int:
         runtime = ManagerRuntime(client)
     except Exception as e:
         log.error("unable to start manager", error=str(e))
+        server.should_exit = True
         return 1
     ready["ok"] = True
     log.info("starting manager")
