@@ -15,6 +15,12 @@ The controller is agnostic: it only sees the protocol
 (wva_amd/kube/client.py), so the in-memory fake and this client are
 interchangeable (exercised by tests/test_kube_http.py against a stub API
 server).
+
+Note on events: this client does not implement watches, so a
+ManagerRuntime over it is purely RequeueAfter-driven — which is also the
+reference's steady state (its Create-only event filter makes creation a
+one-shot trigger; everything else is the timer,
+variantautoscaling_controller.go:473-486).
 """
 
 from __future__ import annotations
