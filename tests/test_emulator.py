@@ -165,8 +165,15 @@ class TestHTTPServer:
             body = resp.json()
             assert body["usage"]["prompt_tokens"] == 3
             assert body["usage"]["completion_tokens"] == 5
+            comp = httpx.post(
+                f"{base}/v1/completions",
+                json={"model": "http-model", "prompt": "one two three four"},
+                timeout=30.0,
+            )
+            assert comp.status_code == 200
+            assert comp.json()["usage"]["prompt_tokens"] == 4
             metrics_text = httpx.get(f"{base}/metrics", timeout=5.0).text
-            assert 'vllm:request_success_total{model_name="http-model"} 1.0' in metrics_text
+            assert 'vllm:request_success_total{model_name="http-model"} 2.0' in metrics_text
             assert "vllm:time_to_first_token_seconds_sum" in metrics_text
             assert "vllm:request_prompt_tokens_sum" in metrics_text
         finally:

@@ -164,3 +164,28 @@ class TestPromlibAPI:
         api = PromlibAPI(make_store())
         with pytest.raises(PromQueryError):
             api.query("sum(")
+
+
+class TestScraperLoop:
+    def test_background_interval_scraping(self):
+        import time
+
+        store = make_store()
+        scraper = Scraper(store)
+        calls = {"n": 0}
+
+        def fetch():
+            calls["n"] += 1
+            return f"loop_metric {calls['n']}\n"
+
+        scraper.add_target(fetch)
+        scraper.start(interval=0.05)
+        time.sleep(0.4)
+        scraper.stop()
+        assert calls["n"] >= 3  # several periodic scrapes happened
+        (series,) = store.select("loop_metric", {})
+        assert len(series.samples) == calls["n"]
+        # stop() joins the thread: no further scrapes
+        n = calls["n"]
+        time.sleep(0.15)
+        assert calls["n"] == n

@@ -35,6 +35,14 @@ class ChatCompletionRequest(BaseModel):
     stream: Optional[bool] = False
 
 
+class CompletionRequest(BaseModel):
+    model: str = "emulated-model"
+    prompt: str = ""
+    max_tokens: Optional[int] = 512
+    temperature: Optional[float] = 0.1
+    stream: Optional[bool] = False
+
+
 class OutputLengthSampler:
     def __init__(self, avg_generated: int, distribution: str) -> None:
         self.avg = avg_generated
@@ -95,6 +103,28 @@ def create_app(settings: Optional[EmulatorSettings] = None) -> FastAPI:
                     },
                 }
             ],
+            "usage": {
+                "prompt_tokens": req.input_tokens,
+                "completion_tokens": req.generated,
+                "total_tokens": req.token_len,
+            },
+        }
+
+    @app.post("/v1/completions")
+    async def completions(request: CompletionRequest):
+        input_len = max(len(request.prompt.split()), 1)
+        req = RequestElement(
+            req_id=str(uuid.uuid4()),
+            input_tokens=input_len,
+            output_tokens=input_len + sampler.sample(),
+        )
+        await engine.submit_and_wait(req)
+        return {
+            "id": req.req_id,
+            "object": "text_completion",
+            "created": int(time.time()),
+            "model": request.model,
+            "choices": [{"index": 0, "text": f"emulated {req.generated} tokens", "finish_reason": "stop"}],
             "usage": {
                 "prompt_tokens": req.input_tokens,
                 "completion_tokens": req.generated,
