@@ -524,3 +524,21 @@ class TestModelAnalyzerResponse:
         va = api.VariantAutoscaling(metadata=ObjectMeta(name="x", namespace="y"))
         response = ModelAnalyzer(System()).analyze_model(va)
         assert response.allocations == {}
+
+
+class TestOptimizationFailurePath:
+    def test_infeasible_everywhere_sets_optimization_failed(self, cluster, prom, registry):
+        """A prepared variant with no feasible allocation anywhere drives
+        the optimizer-failure fanout: OptimizationReady=False persisted,
+        cycle requeues (controller.go:168-186)."""
+        make_deployment(cluster)
+        # alpha far above the Premium slo-tpot=24 on every accelerator
+        make_va(cluster, alpha="500.0", beta="5.0")
+        set_load_metrics(prom, "default/llama-8b", "default", arrival_rps=5.0)
+        result = VariantAutoscalingReconciler(cluster, prom).reconcile()
+        assert result.requeue_after == 1.0  # still requeues
+        va = get_va(cluster)
+        cond = v1alpha1.get_condition(va, v1alpha1.TYPE_OPTIMIZATION_READY)
+        assert cond is not None and cond.status == "False"
+        assert cond.reason == v1alpha1.REASON_OPTIMIZATION_FAILED
+        assert va.status.desired_optimized_alloc.accelerator == ""
