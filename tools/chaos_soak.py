@@ -1,0 +1,141 @@
+"""Failure-injection soak: kill the model server mid-run and verify the
+controller's graceful degradation live.
+
+Timeline: load -> emulator outage past the 5-minute staleness gate ->
+recovery.  Expected behavior (the reference's contract, preserved here):
+during the outage the variant flips to MetricsStale and is *skipped* —
+its last good decision stays in place (no reset, no scale-to-zero); on
+recovery it resumes tracking.
+
+    python tools/chaos_soak.py --out chaos.json
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import sys
+import time
+from pathlib import Path
+
+ROOT = Path(__file__).resolve().parent.parent
+sys.path.insert(0, str(ROOT))
+sys.path.insert(0, str(ROOT / "tools"))
+sys.path.insert(0, str(ROOT / "tests"))
+
+
+def main() -> None:
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--outage-seconds", type=float, default=330.0, help="must exceed the 5-min staleness gate")
+    ap.add_argument("--out", default="")
+    args = ap.parse_args()
+
+    import logging
+    import os
+
+    os.environ.setdefault("WVA_RATE_WINDOW", "8s")
+
+    import torch
+
+    from wva_amd.controller.logger import log as wva_log
+
+    wva_log.init(logging.ERROR)
+
+    from prometheus_client import CollectorRegistry
+
+    from vllm_emulator.engine import EmulatorSettings
+    from wva_amd.api import v1alpha1
+    from wva_amd.controller import metrics as ctrl_metrics
+    from wva_amd.controller.reconciler import VariantAutoscalingReconciler
+    from wva_amd.promlib import PromlibAPI, Scraper, TimeSeriesStore
+    from kube_fixtures import make_cluster, make_deployment, make_va
+    from test_e2e import EmulatorProcess, drive_load, simulate_hpa
+
+    registry = CollectorRegistry()
+    ctrl_metrics.init_metrics(registry)
+    device = "cuda" if torch.cuda.is_available() else "cpu"
+
+    cluster = make_cluster(opt_interval="1s")
+    make_deployment(cluster, replicas=1)
+    make_va(cluster, max_batch=16, alpha="12.0", beta="6.0", gamma="4.0", delta="0.01")
+    store = TimeSeriesStore()
+    scraper = Scraper(store)
+    rec = VariantAutoscalingReconciler(
+        cluster, PromlibAPI(store),
+        batched_analyzer=device == "cuda",
+        analyzer_device=device if device == "cuda" else None,
+    )
+
+    settings = EmulatorSettings(
+        model="default/llama-8b", decode_alpha=12.0, decode_beta=6.0,
+        prefill_gamma=4.0, prefill_delta=0.01, avg_generated_len=25,
+        tokens_distribution="deterministic", max_batch_size=16, realtime=True,
+    )
+
+    def checkpoint(phase):
+        rec.reconcile()
+        va = cluster.get(v1alpha1.VariantAutoscaling, "vllm-llama", "default")
+        cond = v1alpha1.get_condition(va, v1alpha1.TYPE_METRICS_AVAILABLE)
+        simulate_hpa(cluster, registry)
+        entry = {
+            "phase": phase,
+            "desired": va.status.desired_optimized_alloc.num_replicas,
+            "metrics_status": cond.status if cond else "unset",
+            "metrics_reason": cond.reason if cond else "",
+        }
+        timeline.append(entry)
+        print(json.dumps(entry))
+        return entry
+
+    timeline = []
+    # phase 1: healthy under load
+    with EmulatorProcess(settings) as emu:
+        scraper.add_target(f"{emu.base_url}/metrics", extra_labels={"namespace": "default"})
+        scraper.start(interval=0.5)
+        try:
+            drive_load(emu.base_url, rate_rps=5.0, duration_s=8.0)
+            healthy = checkpoint("healthy-load")
+        finally:
+            scraper.stop()
+    # emulator context exited: the "server" is down; the scraper keeps
+    # failing (stopped here; samples age in the store)
+    peak = healthy["desired"]
+
+    # phase 2: outage — before the staleness gate the series are merely
+    # flat (arrival decays to 0 at the window edge, desired decays to min);
+    # after 5 minutes they are STALE and the variant is skipped entirely
+    time.sleep(min(30.0, args.outage_seconds))
+    early = checkpoint("outage-early")
+    remaining = args.outage_seconds - 30.0
+    if remaining > 0:
+        time.sleep(remaining)
+    stale = checkpoint("outage-stale")
+
+    # phase 3: recovery — a fresh emulator at the same model name
+    with EmulatorProcess(settings) as emu:
+        scraper2 = Scraper(store)
+        scraper2.add_target(f"{emu.base_url}/metrics", extra_labels={"namespace": "default"})
+        scraper2.start(interval=0.5)
+        try:
+            drive_load(emu.base_url, rate_rps=5.0, duration_s=8.0)
+            recovered = checkpoint("recovered")
+        finally:
+            scraper2.stop()
+
+    result = {
+        "timeline": timeline,
+        "peak": peak,
+        # the stale phase must SKIP (status False/Stale) and hold the last
+        # decision rather than resetting it
+        "stale_skipped": stale["metrics_reason"] == "MetricsStale",
+        "decision_held_through_outage": stale["desired"] == early["desired"],
+        "recovered_tracking": recovered["metrics_status"] == "True" and recovered["desired"] >= 1,
+    }
+    print(json.dumps({k: v for k, v in result.items() if k != "timeline"}))
+    if args.out:
+        with open(args.out, "w") as f:
+            json.dump(result, f, indent=2)
+
+
+if __name__ == "__main__":
+    main()
