@@ -194,6 +194,7 @@ def config_5():
     import os
 
     os.environ.setdefault("WVA_RATE_WINDOW", "8s")
+    prev_scale_to_zero = os.environ.get("WVA_SCALE_TO_ZERO")
     os.environ["WVA_SCALE_TO_ZERO"] = "true"
     from prometheus_client import CollectorRegistry
 
@@ -242,6 +243,10 @@ def config_5():
                     simulate_hpa(cluster, registry)
             finally:
                 scraper.stop()
+        if prev_scale_to_zero is None:
+            os.environ.pop("WVA_SCALE_TO_ZERO", None)
+        else:
+            os.environ["WVA_SCALE_TO_ZERO"] = prev_scale_to_zero
         ramp_ok = trajectory[0] == 0 and trajectory[1] >= 1 and trajectory[2] == 0
         return {
             "solver_wall_clock_ms": float(np.mean(cycle_ms)),
