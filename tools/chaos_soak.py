@@ -72,7 +72,15 @@ def main() -> None:
         tokens_distribution="deterministic", max_batch_size=16, realtime=True,
     )
 
+    from wva_amd.controller import collector
+
     def checkpoint(phase):
+        # the availability gate as the cycle sees it (a skipped variant's
+        # persisted conditions intentionally KEEP their last good values,
+        # so the gate must be observed directly)
+        gate = collector.validate_metrics_availability(
+            rec.prom_api, "default/llama-8b", "default"
+        )
         rec.reconcile()
         va = cluster.get(v1alpha1.VariantAutoscaling, "vllm-llama", "default")
         cond = v1alpha1.get_condition(va, v1alpha1.TYPE_METRICS_AVAILABLE)
@@ -80,8 +88,10 @@ def main() -> None:
         entry = {
             "phase": phase,
             "desired": va.status.desired_optimized_alloc.num_replicas,
-            "metrics_status": cond.status if cond else "unset",
-            "metrics_reason": cond.reason if cond else "",
+            "gate_available": gate.available,
+            "gate_reason": gate.reason,
+            "persisted_status": cond.status if cond else "unset",
+            "persisted_reason": cond.reason if cond else "",
         }
         timeline.append(entry)
         print(json.dumps(entry))
@@ -125,11 +135,14 @@ def main() -> None:
     result = {
         "timeline": timeline,
         "peak": peak,
-        # the stale phase must SKIP (status False/Stale) and hold the last
-        # decision rather than resetting it
-        "stale_skipped": stale["metrics_reason"] == "MetricsStale",
+        # past the 5-minute lookback the instant vectors are EMPTY (both
+        # for promlib and real Prometheus), so the gate trips as
+        # MetricsMissing/MetricsStale and the cycle SKIPS the variant:
+        # the persisted conditions and the last decision stay untouched
+        "outage_gate_tripped": not stale["gate_available"],
+        "skip_preserved_conditions": stale["persisted_status"] == "True",
         "decision_held_through_outage": stale["desired"] == early["desired"],
-        "recovered_tracking": recovered["metrics_status"] == "True" and recovered["desired"] >= 1,
+        "recovered_tracking": recovered["gate_available"] and recovered["desired"] >= 1,
     }
     print(json.dumps({k: v for k, v in result.items() if k != "timeline"}))
     if args.out:
