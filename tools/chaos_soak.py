@@ -1,11 +1,13 @@
 """Failure-injection soak: kill the model server mid-run and verify the
 controller's graceful degradation live.
 
-Timeline: load -> emulator outage past the 5-minute staleness gate ->
-recovery.  Expected behavior (the reference's contract, preserved here):
-during the outage the variant flips to MetricsStale and is *skipped* —
-its last good decision stays in place (no reset, no scale-to-zero); on
-recovery it resumes tracking.
+Timeline: load -> emulator outage past the 5-minute lookback/staleness
+horizon -> recovery.  Expected behavior (the reference's contract,
+preserved here): within the rate window the flat counters read as zero
+arrivals (decay to the minimum); past the horizon the instant vectors go
+empty, the availability gate trips (MetricsMissing/MetricsStale) and the
+variant is *skipped* — persisted conditions and the last decision stay
+untouched; on recovery it resumes tracking.
 
     python tools/chaos_soak.py --out chaos.json
 """
