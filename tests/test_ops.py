@@ -75,6 +75,17 @@ class TestNativeCPUParity:
         assert_results_close(got, want, rtol=1e-4)
         assert got[:, R_FEASIBLE].sum() > 0  # exercise the feasible path
 
+    def test_large_n_parity(self):
+        # beyond GPU_MAX_BATCH_LIMIT: the CPU windowed sweep handles
+        # K = 11*N state chains of ~16k states (the size the LDS-limit
+        # fallback in batched.py routes to this path on GPU boxes)
+        problems = random_problems(12, max_batch_hi=1500, seed=17)
+        problems[0, 6] = 1500.0  # pin one at the top of the range
+        got = solve_problems(problems, device="cpu")
+        want = _solve_problems_python(problems)
+        assert_results_close(got, want, rtol=1e-4)
+        assert got[:, R_FEASIBLE].sum() > 0
+
     def test_infeasible_targets(self):
         problems = random_problems(4, seed=14)
         problems[:, 8] = 0.01  # ITL target below alpha: infeasible
