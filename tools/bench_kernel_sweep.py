@@ -1,0 +1,85 @@
+"""Kernel geometry sweep: single-wave (64) vs 4-wave (256) workgroups
+across max-batch sizes N up to the LDS-resident limit.
+
+The fleet-size benches (tools/bench_queue_solver.py) showed the
+single-wave kernel 20-31% faster at typical N (<=256).  Larger N means a
+longer cumulative-table build (K = 11N states) and a wider significant
+window per model evaluation, which could favor the extra parallelism of
+4 waves — this sweep measures where (if anywhere) the crossover sits.
+
+    python tools/bench_kernel_sweep.py --out gpurun_out/kernel_sweep.json
+"""
+
+import argparse
+import json
+import os
+import sys
+import time
+from pathlib import Path
+
+sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
+sys.path.insert(0, str(Path(__file__).resolve().parent.parent / "tests"))
+
+import numpy as np
+
+from wva_amd.ops import get_native, solve_problems
+from wva_amd.ops.batched import P_MAX_BATCH, R_FEASIBLE
+from test_ops import random_problems
+
+
+def timeit(fn, warmup=3, iters=10):
+    for _ in range(warmup):
+        fn()
+    t0 = time.perf_counter()
+    for _ in range(iters):
+        fn()
+    return (time.perf_counter() - t0) / iters
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--batch", type=int, default=4096)
+    ap.add_argument("--max-batches", type=int, nargs="+", default=[64, 128, 256, 512, 700])
+    ap.add_argument("--out", default="")
+    args = ap.parse_args()
+
+    import torch
+
+    assert torch.cuda.is_available(), "sweep needs an MI355X"
+    native = get_native()
+    rows = []
+    for n_hi in args.max_batches:
+        problems = random_problems(args.batch, max_batch_hi=n_hi, seed=31)
+        problems[:, P_MAX_BATCH] = float(n_hi)  # pin N: uniform K = 11N chains
+        t = torch.from_numpy(problems).cuda()
+        entry = {"batch": args.batch, "N": n_hi, "K": 11 * n_hi}
+        for threads in (64, 256):
+            os.environ["WVA_GPU_THREADS"] = str(threads)
+
+            def run():
+                native.solve_allocations(t)
+                torch.cuda.synchronize()
+
+            entry[f"gpu{threads}_ms"] = timeit(run) * 1e3
+        os.environ.pop("WVA_GPU_THREADS", None)
+        # parity guard: both geometries and the CPU path must agree
+        os.environ["WVA_GPU_THREADS"] = "256"
+        g256 = native.solve_allocations(t).cpu().numpy()
+        os.environ.pop("WVA_GPU_THREADS", None)
+        g64 = native.solve_allocations(t).cpu().numpy()
+        cpu = solve_problems(problems, device="cpu")
+        entry["geom_feas_agree"] = bool((g64[:, R_FEASIBLE] == g256[:, R_FEASIBLE]).all())
+        entry["cpu_feas_flips"] = int((g64[:, R_FEASIBLE] != cpu[:, R_FEASIBLE]).sum())
+        entry["feasible_frac"] = float(g64[:, R_FEASIBLE].mean())
+        entry["speedup_64_over_256"] = entry["gpu256_ms"] / entry["gpu64_ms"]
+        rows.append(entry)
+        print(json.dumps(entry))
+
+    if args.out:
+        Path(args.out).parent.mkdir(parents=True, exist_ok=True)
+        with open(args.out, "w") as f:
+            json.dump(rows, f, indent=2)
+
+
+if __name__ == "__main__":
+    main()

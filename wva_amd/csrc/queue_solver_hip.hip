@@ -38,7 +38,8 @@ namespace wva {
 //   - 64 threads (1 wave64): barrier-free — the wave executes in lockstep,
 //     reductions are pure __shfl_down chains and the combined values are
 //     broadcast from lane 0 with __shfl (no LDS round-trip).
-// The launcher picks by the WVA_GPU_THREADS env var (default 256).
+// The launcher defaults to the single-wave kernel; WVA_GPU_THREADS=256
+// selects the 4-wave geometry (read per launch).
 
 // Reduction scratch layout (doubles, after cum[max_k] in dynamic LDS):
 //   red[0..WAVES*5-1]  per-wave partials (S, Ni, Snum, Ninum, eK)
