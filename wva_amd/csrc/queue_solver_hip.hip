@@ -38,8 +38,9 @@ namespace wva {
 //   - 64 threads (1 wave64): barrier-free — the wave executes in lockstep,
 //     reductions are pure __shfl_down chains and the combined values are
 //     broadcast from lane 0 with __shfl (no LDS round-trip).
-// The launcher defaults to the single-wave kernel; WVA_GPU_THREADS=256
-// selects the 4-wave geometry (read per launch).
+// The launcher auto-selects by state-chain length (single wave below
+// max_k 4096, 4-wave above — the measured crossover); the
+// WVA_GPU_THREADS env var (64|256) overrides, read per launch.
 
 // Reduction scratch layout (doubles, after cum[max_k] in dynamic LDS):
 //   red[0..WAVES*5-1]  per-wave partials (S, Ni, Snum, Ninum, eK)
@@ -299,10 +300,20 @@ extern "C" __global__ void __launch_bounds__(64) wva_solve_kernel_64(
 
 extern "C" void wva_launch_solve(const double *prob, double *out, int n_problems,
                                  int max_k, void *stream) {
-  // single-wave is the default: measured 20-31% faster than the 4-wave
-  // geometry at fleet batch sizes (profiles/r01_queue_solver.md)
+  // geometry auto-selects on state-chain length: the barrier-free single
+  // wave wins 1.2-1.9x up to K~2816 (N<=256) but the 4-wave kernel's
+  // extra sweep parallelism wins 1.2-1.34x from K~5632 (N>=512) — the
+  // measured crossover sits between (profiles/r01_queue_solver.md,
+  // profiles/r01_kernel_sweep.json).  WVA_GPU_THREADS=64|256 overrides.
   const char *env = std::getenv("WVA_GPU_THREADS");
-  const bool one_wave = env == nullptr || std::strcmp(env, "256") != 0;
+  bool one_wave;
+  if (env != nullptr && std::strcmp(env, "256") == 0) {
+    one_wave = false;
+  } else if (env != nullptr && std::strcmp(env, "64") == 0) {
+    one_wave = true;
+  } else {
+    one_wave = max_k < 4096;
+  }
   const int threads = one_wave ? 64 : 256;
   const size_t smem = (size_t)(max_k + threads + 32) * sizeof(double);
   if (one_wave) {
