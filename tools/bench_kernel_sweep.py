@@ -53,15 +53,17 @@ def main():
         problems[:, P_MAX_BATCH] = float(n_hi)  # pin N: uniform K = 11N chains
         t = torch.from_numpy(problems).cuda()
         entry = {"batch": args.batch, "N": n_hi, "K": 11 * n_hi}
-        for threads in (64, 256):
-            os.environ["WVA_GPU_THREADS"] = str(threads)
+        for threads in (64, 256, None):
+            if threads is None:
+                os.environ.pop("WVA_GPU_THREADS", None)  # launcher auto-select
+            else:
+                os.environ["WVA_GPU_THREADS"] = str(threads)
 
             def run():
                 native.solve_allocations(t)
                 torch.cuda.synchronize()
 
-            entry[f"gpu{threads}_ms"] = timeit(run) * 1e3
-        os.environ.pop("WVA_GPU_THREADS", None)
+            entry[f"gpu{threads or 'auto'}_ms"] = timeit(run) * 1e3
         # parity guard: both geometries and the CPU path must agree
         os.environ["WVA_GPU_THREADS"] = "256"
         g256 = native.solve_allocations(t).cpu().numpy()
