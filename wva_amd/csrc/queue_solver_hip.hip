@@ -33,14 +33,19 @@
 
 namespace wva {
 
-// Two workgroup geometries are instantiated:
+// Three workgroup geometries are instantiated:
 //   - 256 threads (4 wave64): strided sweeps + LDS cross-wave combine;
 //   - 64 threads (1 wave64): barrier-free — the wave executes in lockstep,
 //     reductions are pure __shfl_down chains and the combined values are
-//     broadcast from lane 0 with __shfl (no LDS round-trip).
-// The launcher auto-selects by state-chain length (single wave below
-// max_k 4096, 4-wave above — the measured crossover); the
-// WVA_GPU_THREADS env var (64|256) overrides, read per launch.
+//     broadcast from lane 0 with __shfl (no LDS round-trip);
+//   - 128 threads (2 wave64, "dual"): wave 0 runs the TTFT bisection
+//     while wave 1 concurrently runs the ITL bisection, each with its own
+//     barrier-free 64-lane evaluations — for the common both-targets
+//     case this halves the serial-bisection critical path, which is what
+//     bounds small-fleet dispatches (a 192-problem launch cannot fill
+//     256 CUs, so per-problem latency IS the dispatch time).
+// The launcher auto-selects by state-chain length; the WVA_GPU_THREADS
+// env var (64|128|256) overrides, read per launch.
 
 // Reduction scratch layout (doubles, after cum[max_k] in dynamic LDS):
 //   red[0..WAVES*5-1]  per-wave partials (S, Ni, Snum, Ninum, eK)
@@ -76,10 +81,14 @@ struct WgEval {
 
     // single windowed pass: normalization and moment sums (eK is the
     // state-K boundary term, reduced like the sums — only its owner
-    // thread contributes a non-zero partial)
+    // thread contributes a non-zero partial).  For WAVES == 1 the index
+    // is the LANE, not the block thread id: that makes WgEval<64> usable
+    // per-wave inside the 128-thread dual kernel (each wave sweeps its
+    // own full window in lockstep, reductions stay wave-local).
     const int num = p.max_batch;
+    const int idx = (WAVES == 1) ? lane : tid;
     double S = 0.0, Ni = 0.0, Snum = 0.0, Ninum = 0.0, eK = 0.0;
-    for (int n = n_lo + tid; n <= n_hi; n += THREADS) {
+    for (int n = n_lo + idx; n <= n_hi; n += THREADS) {
       double e = exp(log_p(cum, loglam, n) - m);
       S += e;
       Ni += (double)n * e;
@@ -190,15 +199,7 @@ __device__ int wg_binary_search(double x_min, double x_max, double y_target, F e
   return 0;
 }
 
-template <int THREADS>
-__device__ void solve_body(const double *__restrict__ prob, double *__restrict__ out,
-                           int n_problems, int max_k) {
-  const int pid = blockIdx.x;
-  if (pid >= n_problems) return;
-  const double *pr = prob + (size_t)pid * PROBLEM_FIELDS;
-  double *res = out + (size_t)pid * RESULT_FIELDS;
-  const int tid = threadIdx.x;
-
+__device__ inline Parms load_parms(const double *pr) {
   Parms p;
   p.alpha = pr[P_ALPHA];
   p.beta = pr[P_BETA];
@@ -209,15 +210,13 @@ __device__ void solve_body(const double *__restrict__ prob, double *__restrict__
   p.max_batch = (int)pr[P_MAX_BATCH];
   p.num_decode = p.out_tokens - 1;
   if (p.in_tokens == 0.0 && p.out_tokens == 1) p.num_decode = 1;
+  return p;
+}
 
-  const int K = p.max_batch * (1 + kMaxQueueToBatchRatio);
-
-  extern __shared__ double smem[];
-  double *cum = smem;             // this problem's K entries
-  double *totals = smem + max_k;  // THREADS chunk totals
-  double *red = totals + THREADS;
-
-  // chunked parallel inclusive scan of log_mu over K states
+// chunked parallel inclusive scan of log_mu over K states into LDS
+template <int THREADS>
+__device__ void build_cum(const Parms &p, int K, double *cum, double *totals) {
+  const int tid = threadIdx.x;
   const int chunk = (K + THREADS - 1) / THREADS;
   const int lo = tid * chunk;
   const int hi = min(lo + chunk, K);
@@ -232,6 +231,26 @@ __device__ void solve_body(const double *__restrict__ prob, double *__restrict__
   for (int t = 0; t < tid; ++t) offset += totals[t];  // broadcast LDS reads
   for (int n = lo; n < hi; ++n) cum[n] += offset;
   __syncthreads();
+}
+
+template <int THREADS>
+__device__ void solve_body(const double *__restrict__ prob, double *__restrict__ out,
+                           int n_problems, int max_k) {
+  const int pid = blockIdx.x;
+  if (pid >= n_problems) return;
+  const double *pr = prob + (size_t)pid * PROBLEM_FIELDS;
+  double *res = out + (size_t)pid * RESULT_FIELDS;
+  const int tid = threadIdx.x;
+
+  const Parms p = load_parms(pr);
+  const int K = p.max_batch * (1 + kMaxQueueToBatchRatio);
+
+  extern __shared__ double smem[];
+  double *cum = smem;             // this problem's K entries
+  double *totals = smem + max_k;  // THREADS chunk totals
+  double *red = totals + THREADS;
+
+  build_cum<THREADS>(p, K, cum, totals);
 
   const double lam_min = serv_rate(p, 1) * kEpsilon;  // req/ms
   const double lam_max = serv_rate(p, p.max_batch) * (1.0 - kEpsilon);
@@ -281,6 +300,92 @@ __device__ void solve_body(const double *__restrict__ prob, double *__restrict__
   }
 }
 
+// 2-wave "dual" body: the TTFT and ITL bisections are independent, so
+// wave 0 searches the TTFT target while wave 1 searches the ITL target
+// concurrently.  Each wave's evaluations are the barrier-free WgEval<64>
+// (lane-indexed, __shfl-only), so the waves may diverge arbitrarily
+// between the scan barrier and the exchange barrier.  Semantics are
+// identical to the sequential body: the original runs TTFT first and
+// returns infeasible without touching ITL — here both run, but the
+// combined feasibility check and the zeroed result row give the same
+// observable output.
+__device__ void solve_body_dual(const double *__restrict__ prob, double *__restrict__ out,
+                                int n_problems, int max_k) {
+  constexpr int THREADS = 128;
+  const int pid = blockIdx.x;
+  if (pid >= n_problems) return;
+  const double *pr = prob + (size_t)pid * PROBLEM_FIELDS;
+  double *res = out + (size_t)pid * RESULT_FIELDS;
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+
+  const Parms p = load_parms(pr);
+  const int K = p.max_batch * (1 + kMaxQueueToBatchRatio);
+
+  extern __shared__ double smem[];
+  double *cum = smem;             // this problem's K entries
+  double *totals = smem + max_k;  // THREADS chunk totals
+  double *ex = totals + THREADS;  // exchange: lam_ttft, ind_ttft, lam_itl, ind_itl
+
+  build_cum<THREADS>(p, K, cum, totals);
+
+  const double lam_min = serv_rate(p, 1) * kEpsilon;  // req/ms
+  const double lam_max = serv_rate(p, p.max_batch) * (1.0 - kEpsilon);
+
+  WgEval<64> ev{p, cum, nullptr, K};  // lane-indexed, barrier-free
+
+  if (tid < RESULT_FIELDS) res[tid] = 0.0;
+
+  double lam_t = lam_max;
+  int ind = 0;
+  if (wave == 0) {
+    if (pr[P_TARGET_TTFT] > 0.0) {
+      ind = wg_binary_search(
+          lam_min, lam_max, pr[P_TARGET_TTFT],
+          [&](double x) { return ev.eval_ttft(x); }, &lam_t);
+    }
+  } else {
+    if (pr[P_TARGET_ITL] > 0.0) {
+      ind = wg_binary_search(
+          lam_min, lam_max, pr[P_TARGET_ITL],
+          [&](double x) { return ev.eval_itl(x); }, &lam_t);
+    }
+  }
+  if ((tid & 63) == 0) {
+    ex[wave * 2 + 0] = lam_t;
+    ex[wave * 2 + 1] = (double)ind;
+  }
+  __syncthreads();
+  if (ex[1] < 0.0 || ex[3] < 0.0) return;  // a target below the reachable range
+  if (wave == 1) return;                   // wave 0 finishes the tail alone
+
+  double lam_tps = lam_max;
+  if (pr[P_TARGET_TPS] > 0.0) lam_tps = lam_max * (1.0 - kStabilityFraction);
+
+  const double lam = fmin(ex[0], fmin(ex[2], lam_tps));
+  Stats st = ev.eval(lam);
+  const double rate_star = st.throughput * 1000.0;  // req/s
+
+  const double total_rate = pr[P_TOTAL_RATE];
+  double n_rep = ceil(total_rate / rate_star);
+  if (n_rep < pr[P_MIN_REPLICAS]) n_rep = pr[P_MIN_REPLICAS];
+  const double rate = total_rate / n_rep;
+  if (rate <= 0.0 || rate > lam_max * 1000.0) return;
+
+  Stats fin = ev.eval(rate / 1000.0);
+  double rho = fin.n_serv / (double)p.max_batch;
+  rho = fmin(fmax(rho, 0.0), 1.0);
+
+  if (tid == 0) {
+    res[R_FEASIBLE] = 1.0;
+    res[R_REPLICAS] = n_rep;
+    res[R_RATE_STAR] = rate_star;
+    res[R_ITL] = eval_itl_of(p, fin);
+    res[R_TTFT] = eval_ttft_of(p, fin);
+    res[R_RHO] = rho;
+  }
+}
+
 extern "C" __global__ void __launch_bounds__(256) wva_solve_kernel_256(
     const double *__restrict__ prob, double *__restrict__ out, int n_problems,
     int max_k) {
@@ -293,6 +398,12 @@ extern "C" __global__ void __launch_bounds__(64) wva_solve_kernel_64(
   solve_body<64>(prob, out, n_problems, max_k);
 }
 
+extern "C" __global__ void __launch_bounds__(128) wva_solve_kernel_128(
+    const double *__restrict__ prob, double *__restrict__ out, int n_problems,
+    int max_k) {
+  solve_body_dual(prob, out, n_problems, max_k);
+}
+
 }  // namespace wva
 
 #include <cstdlib>
@@ -300,24 +411,29 @@ extern "C" __global__ void __launch_bounds__(64) wva_solve_kernel_64(
 
 extern "C" void wva_launch_solve(const double *prob, double *out, int n_problems,
                                  int max_k, void *stream) {
-  // geometry auto-selects on state-chain length: the barrier-free single
-  // wave wins 1.2-1.9x up to K~2816 (N<=256) but the 4-wave kernel's
-  // extra sweep parallelism wins 1.2-1.34x from K~5632 (N>=512) — the
-  // measured crossover sits between (profiles/r01_queue_solver.md,
-  // profiles/r01_kernel_sweep.json).  WVA_GPU_THREADS=64|256 overrides.
+  // Geometry auto-selects on state-chain length (measured:
+  // profiles/r01_queue_solver.md, profiles/r01_kernel_sweep.json).  The
+  // 2-wave dual kernel overlaps the TTFT and ITL bisections and is the
+  // default below the large-K regime; the 4-wave kernel's sweep
+  // parallelism takes over for long chains.  WVA_GPU_THREADS=64|128|256
+  // overrides (64 = the sequential single-wave kernel).
   const char *env = std::getenv("WVA_GPU_THREADS");
-  bool one_wave;
+  int threads;
   if (env != nullptr && std::strcmp(env, "256") == 0) {
-    one_wave = false;
+    threads = 256;
+  } else if (env != nullptr && std::strcmp(env, "128") == 0) {
+    threads = 128;
   } else if (env != nullptr && std::strcmp(env, "64") == 0) {
-    one_wave = true;
+    threads = 64;
   } else {
-    one_wave = max_k < 4096;
+    threads = max_k < 4096 ? 128 : 256;
   }
-  const int threads = one_wave ? 64 : 256;
   const size_t smem = (size_t)(max_k + threads + 32) * sizeof(double);
-  if (one_wave) {
+  if (threads == 64) {
     hipLaunchKernelGGL(wva::wva_solve_kernel_64, dim3(n_problems), dim3(64), smem,
+                       (hipStream_t)stream, prob, out, n_problems, max_k);
+  } else if (threads == 128) {
+    hipLaunchKernelGGL(wva::wva_solve_kernel_128, dim3(n_problems), dim3(128), smem,
                        (hipStream_t)stream, prob, out, n_problems, max_k);
   } else {
     hipLaunchKernelGGL(wva::wva_solve_kernel_256, dim3(n_problems), dim3(256), smem,
