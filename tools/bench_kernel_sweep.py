@@ -67,15 +67,21 @@ def main():
 
                 entry[f"gpu{threads or 'auto'}_ms"] = timeit(run) * 1e3
                 results[threads] = native.solve_allocations(t).cpu().numpy()
-            # parity guards: the dual kernel runs the same per-wave math in
-            # the same order as the single wave -> bit-identical; the 4-wave
-            # kernel's different summation order allows ulp-level drift, so
-            # it is held to feasibility agreement
+            # parity guards: geometries may differ at ulp level (separate
+            # template instantiations contract FMAs differently), so the
+            # gate is feasibility agreement + tight relative closeness of
+            # the rate column on feasible rows
             cpu = solve_problems(problems, device="cpu")
             base = results[64]
-            entry["dual_bitwise_identical"] = bool((results[128] == base).all())
             entry["geom_feas_agree"] = all(
                 (results[g][:, R_FEASIBLE] == base[:, R_FEASIBLE]).all()
+                for g in (128, 256, None)
+            )
+            feas = base[:, R_FEASIBLE] == 1.0
+            entry["max_rel_rate_diff"] = max(
+                float(np.abs(
+                    (results[g][feas, 2] - base[feas, 2]) / np.maximum(base[feas, 2], 1e-300)
+                ).max()) if feas.any() else 0.0
                 for g in (128, 256, None)
             )
             entry["cpu_feas_flips"] = int((base[:, R_FEASIBLE] != cpu[:, R_FEASIBLE]).sum())

@@ -425,8 +425,15 @@ extern "C" void wva_launch_solve(const double *prob, double *out, int n_problems
     threads = 128;
   } else if (env != nullptr && std::strcmp(env, "64") == 0) {
     threads = 64;
+  } else if (n_problems >= 512) {
+    // chip-filling launches: the dual kernel won or tied every measured
+    // (B, K) point at B=4096 (incl. K=7700, where it matches 4-wave)
+    threads = 128;
   } else {
-    threads = max_k < 4096 ? 128 : 256;
+    // underfilled launches are per-problem latency-bound: overlapping
+    // the two bisections wins while evaluations are short, but from
+    // K~2816 the 4-wave sweep parallelism matters more (B=192 data)
+    threads = max_k < 2048 ? 128 : 256;
   }
   const size_t smem = (size_t)(max_k + threads + 32) * sizeof(double);
   if (threads == 64) {
