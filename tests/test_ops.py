@@ -151,6 +151,17 @@ class TestGPU:
         ref = _solve_problems_python(problems)
         assert_results_close(gpu, ref, rtol=1e-4)
 
+    @pytest.mark.parametrize("threads", ["64", "128", "256"])
+    def test_each_geometry_matches_cpu(self, threads, monkeypatch):
+        # pin each kernel geometry explicitly (the launcher's auto pick
+        # would otherwise leave two of the three unexercised)
+        monkeypatch.setenv("WVA_GPU_THREADS", threads)
+        problems = random_problems(128, seed=19)
+        gpu = solve_problems(problems, device="cuda")
+        cpu = solve_problems(problems, device="cpu")
+        assert_results_close(gpu, cpu, rtol=1e-4)
+        assert gpu[:, R_FEASIBLE].sum() > 50
+
     def test_large_batch_limit_falls_back(self):
         problems = random_problems(8, seed=15)
         problems[0, 6] = 1024.0  # beyond the LDS-resident limit
