@@ -311,3 +311,87 @@ class TestEffectiveConcurrency:
         rs = RequestSize(avg_input_tokens=0, avg_output_tokens=1)
         assert effective_concurrency(100.0, sp, rs, 8) == 8.0
         assert effective_concurrency(0.0, sp, rs, 8) == 0.0
+
+
+class TestMG1:
+    """M/G/1 extension (wva_amd/analyzer/mg1.py): P-K formula + SCV
+    mapping.  The reference is Markovian-only; these check the exact
+    special cases the approximation must reproduce."""
+
+    def test_mm1_reduction(self):
+        # scv=1 (exponential) must equal the textbook M/M/1 wait
+        # Wq = rho / (mu - lam)
+        lam, mu = 0.6, 1.0
+        from wva_amd.analyzer import pollaczek_khinchine_wait
+
+        w = pollaczek_khinchine_wait(lam, 1.0 / mu, scv=1.0)
+        rho = lam / mu
+        assert w == pytest.approx(rho / (mu - lam))
+
+    def test_md1_half_wait(self):
+        # deterministic service: exactly half the M/M/1 wait
+        from wva_amd.analyzer import pollaczek_khinchine_wait
+
+        w_exp = pollaczek_khinchine_wait(0.8, 1.0, scv=1.0)
+        w_det = pollaczek_khinchine_wait(0.8, 1.0, scv=0.0)
+        assert w_det == pytest.approx(w_exp / 2)
+
+    def test_unstable_raises(self):
+        from wva_amd.analyzer import pollaczek_khinchine_wait
+
+        with pytest.raises(ValueError):
+            pollaczek_khinchine_wait(1.0, 1.0, scv=1.0)
+        with pytest.raises(ValueError):
+            pollaczek_khinchine_wait(0.5, -1.0, scv=1.0)
+
+    def test_zero_load(self):
+        from wva_amd.analyzer import pollaczek_khinchine_wait
+
+        assert pollaczek_khinchine_wait(0.0, 5.0, scv=0.5) == 0.0
+
+    def test_scv_from_tokens_all_variable(self):
+        # no fixed component: service SCV = token SCV
+        from wva_amd.analyzer import service_scv_from_tokens
+
+        assert service_scv_from_tokens(100.0, 0.7, 0.5, fixed_time=0.0) == pytest.approx(0.7)
+
+    def test_scv_from_tokens_damped_by_fixed(self):
+        # fixed part equal to the variable part: damping (1/2)^2
+        from wva_amd.analyzer import service_scv_from_tokens
+
+        scv = service_scv_from_tokens(100.0, 1.0, 0.5, fixed_time=50.0)
+        assert scv == pytest.approx(0.25)
+
+    def test_corrector_identity_and_scaling(self):
+        from wva_amd.analyzer import MG1Corrector
+
+        assert MG1Corrector(1.0).correct(10.0).wait == pytest.approx(10.0)
+        m = MG1Corrector(0.0).correct(10.0)
+        assert m.wait == pytest.approx(5.0)
+        assert m.correction == pytest.approx(0.5)
+        assert m.markovian_wait == 10.0
+        with pytest.raises(ValueError):
+            MG1Corrector(-0.1)
+        with pytest.raises(ValueError):
+            MG1Corrector(1.0).correct(float("nan"))
+
+    def test_consistency_with_statedep_large_n(self):
+        # sanity: for a single-slot exponential server (N=1, huge K) the
+        # state-dependent chain's wait approaches the M/M/1 wait that
+        # scv=1 P-K reproduces — ties the extension to the main analyzer
+        from wva_amd.analyzer import pollaczek_khinchine_wait
+
+        config = Configuration(
+            max_batch_size=1,
+            max_queue_size=2000,
+            service_parms=ServiceParms(
+                prefill=PrefillParms(gamma=0.0, delta=0.0),
+                decode=DecodeParms(alpha=10.0, beta=0.0),
+            ),
+        )
+        qa = QueueAnalyzer(config, RequestSize(avg_input_tokens=0, avg_output_tokens=2))
+        serv_time_ms = 10.0  # one decode pass: (K-1) * alpha
+        lam_per_s = 50.0  # rho = 0.5 (analyze takes req/s; services are ms)
+        metrics = qa.analyze(lam_per_s)
+        pk_ms = pollaczek_khinchine_wait(lam_per_s / 1000.0, serv_time_ms, scv=1.0)
+        assert metrics.avg_wait_time == pytest.approx(pk_ms, rel=0.05)
