@@ -213,3 +213,63 @@ class TestDefaultServiceClass:
         assert create_allocation(system, "s:ns", "MI355X") is None
         # priority falls back to the default lowest
         assert server.priority(system) == 100
+
+
+class TestKVCacheCapacity:
+    """MI355X HBM capacity model (wva_amd/core/kvcache.py)."""
+
+    def _llama8b(self):
+        from wva_amd.core.kvcache import ModelMemoryProfile
+
+        return ModelMemoryProfile.from_architecture(
+            params_billions=8, layers=32, kv_heads=8, head_dim=128
+        )
+
+    def test_kv_bytes_per_token_gqa(self):
+        from wva_amd.core.kvcache import kv_bytes_per_token
+
+        # Llama-3.1-8B GQA: 2 * 32 * 8 * 128 * 2 = 128 KiB
+        assert kv_bytes_per_token(32, 8, 128, 2) == 128 * 1024
+        with pytest.raises(ValueError):
+            kv_bytes_per_token(0, 8, 128)
+
+    def test_mi355x_8b_capacity(self):
+        from wva_amd.core.kvcache import max_batch_for_context, max_concurrent_tokens
+
+        profile = self._llama8b()
+        tokens = max_concurrent_tokens(288, profile)
+        assert 1_900_000 < tokens < 2_100_000  # ~2M tokens beside 16 GB weights
+        assert 450 <= max_batch_for_context(288, profile, 4096) <= 520
+
+    def test_weights_do_not_fit(self):
+        from wva_amd.core.kvcache import ModelMemoryProfile, max_concurrent_tokens
+
+        huge = ModelMemoryProfile.from_architecture(500, 120, 8, 128)  # 1 TB bf16
+        assert max_concurrent_tokens(288, huge) == 0
+
+    def test_fp8_doubles_capacity(self):
+        from wva_amd.core.kvcache import ModelMemoryProfile, max_concurrent_tokens
+
+        bf16 = self._llama8b()
+        fp8 = ModelMemoryProfile.from_architecture(8, 32, 8, 128, dtype_bytes=1)
+        # both weights and KV halve -> more than 2x the token budget
+        assert max_concurrent_tokens(288, fp8) > 2 * max_concurrent_tokens(288, bf16)
+
+    def test_validate_max_batch(self):
+        from wva_amd.core.kvcache import validate_max_batch
+
+        profile = self._llama8b()
+        assert validate_max_batch(256, 288, profile, 4096) == ""
+        warn = validate_max_batch(4096, 288, profile, 4096)
+        assert "exceeds KV capacity" in warn and "4096" in warn
+
+    def test_input_validation(self):
+        from wva_amd.core.kvcache import max_batch_for_context, max_concurrent_tokens
+
+        profile = self._llama8b()
+        with pytest.raises(ValueError):
+            max_concurrent_tokens(0, profile)
+        with pytest.raises(ValueError):
+            max_concurrent_tokens(288, profile, overhead_fraction=1.0)
+        with pytest.raises(ValueError):
+            max_batch_for_context(288, profile, 0)

@@ -160,3 +160,23 @@ class TestMoELayer:
             moe_out = layer(x)
             dense = F.silu(x @ layer.w_up[0]) @ layer.w_down[0]
         assert torch.allclose(moe_out, dense, atol=1e-5)
+
+
+class TestKVPlan:
+    def test_cli_8b_mi355x(self):
+        import json
+        import subprocess
+        import sys
+        from pathlib import Path
+
+        out = subprocess.run(
+            [sys.executable, str(Path(__file__).parent.parent / "tools" / "kv_plan.py"),
+             "--params-b", "8", "--layers", "32", "--kv-heads", "8",
+             "--head-dim", "128", "--context", "4096"],
+            capture_output=True, text=True, timeout=120,
+        )
+        assert out.returncode == 0, out.stderr
+        data = json.loads(out.stdout)
+        assert data["mem_size_gb"] == 288
+        assert data["kv_kib_per_token"] == 128.0
+        assert 450 <= data["max_batch"] <= 520
