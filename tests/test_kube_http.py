@@ -120,3 +120,52 @@ class TestReconcilerOverHTTP:
         assert va.status.desired_optimized_alloc.num_replicas >= 1
         assert v1alpha1.is_condition_true(va, v1alpha1.TYPE_OPTIMIZATION_READY)
         assert any(r.kind == "Deployment" for r in va.metadata.owner_references)
+
+
+class TestWatch:
+    def test_watch_create_streams_new_objects(self, api_server, client):
+        import queue
+
+        from wva_amd.api.v1alpha1.types import ObjectMeta
+
+        _, store = api_server
+        got = queue.Queue()
+
+        def consume():
+            for obj in client.watch_create(
+                v1alpha1.VariantAutoscaling, namespace="default", timeout_seconds=3
+            ):
+                got.put(obj)
+
+        t = threading.Thread(target=consume, daemon=True)
+        t.start()
+        time.sleep(0.3)  # let the watch connect
+        # wrong-type and wrong-namespace creations must NOT be streamed
+        store.create(ConfigMap(metadata=ObjectMeta(name="noise", namespace="default")))
+        make_va(store, name="watched-va")
+        first = got.get(timeout=5.0)
+        assert first.metadata.name == "watched-va"
+        t.join(timeout=8.0)
+        assert not t.is_alive()  # window closed by timeoutSeconds
+        assert got.empty()
+
+    def test_manager_runtime_wakes_on_create_over_http(self, api_server, client, monkeypatch):
+        from wva_amd.controller.reconciler import ManagerRuntime
+
+        _, store = api_server
+        registry = CollectorRegistry()
+        ctrl_metrics.init_metrics(registry)
+        try:
+            runtime = ManagerRuntime(client, prom_api=MockPromAPI())
+            assert runtime._watch_threads  # HTTP tier gets watcher threads
+            assert not runtime._wake.is_set()
+            time.sleep(0.3)
+            make_va(store, name="wake-va")
+            for _ in range(60):
+                if runtime._wake.is_set():
+                    break
+                time.sleep(0.05)
+            assert runtime._wake.is_set()
+            runtime.stop()
+        finally:
+            ctrl_metrics.reset_metrics()

@@ -428,8 +428,20 @@ class ManagerRuntime:
         self.reconciler = VariantAutoscalingReconciler(client, prom_api, **reconciler_kw)
         self._wake = threading.Event()
         self._stop = threading.Event()
+        self._watch_threads: list = []
         if hasattr(client, "on_create"):
             client.on_create(self._on_create)
+        elif hasattr(client, "watch_create"):
+            # HTTP tier: Create-event wakeups via bounded watch windows
+            for cls, namespace in (
+                (v1alpha1.VariantAutoscaling, None),
+                (ConfigMap, CONFIG_MAP_NAMESPACE),
+            ):
+                t = threading.Thread(
+                    target=self._watch_loop, args=(cls, namespace), daemon=True
+                )
+                t.start()
+                self._watch_threads.append(t)
 
     def _get_prometheus_config(self) -> PrometheusConfig:
         config = parse_prometheus_config_from_env()
@@ -456,6 +468,22 @@ class ManagerRuntime:
             "no Prometheus configuration found. Please set PROMETHEUS_BASE_URL "
             "environment variable or configure via ConfigMap"
         )
+
+    def _watch_loop(self, cls, namespace) -> None:
+        """One watch window at a time; reconnect until stopped.  Errors
+        back off briefly — a dead watch degrades to timer-only cadence,
+        never to a crashed runtime."""
+        while not self._stop.is_set():
+            try:
+                for obj in self.client.watch_create(
+                    cls, namespace=namespace, timeout_seconds=5
+                ):
+                    self._on_create(obj)
+                    if self._stop.is_set():
+                        return
+            except Exception as e:
+                log.debug("watch window failed; reconnecting", kind=cls.__name__, error=str(e))
+                self._stop.wait(1.0)
 
     def _on_create(self, obj) -> None:
         # Create-only event filter: VAs and the watched ConfigMap enqueue

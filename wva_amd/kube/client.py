@@ -70,6 +70,13 @@ class InMemoryKubeClient:
         reconciler reacts to)."""
         self._create_hooks.append(hook)
 
+    def remove_create_hook(self, hook: Callable[[object], None]) -> None:
+        """Unsubscribe a hook (watch connections detach on close)."""
+        try:
+            self._create_hooks.remove(hook)
+        except ValueError:
+            pass
+
     # -- CRUD ---------------------------------------------------------------
     def get(self, cls: Type[T], name: str, namespace: str) -> T:
         with self._lock:

@@ -16,11 +16,11 @@ The controller is agnostic: it only sees the protocol
 interchangeable (exercised by tests/test_kube_http.py against a stub API
 server).
 
-Note on events: this client does not implement watches, so a
-ManagerRuntime over it is purely RequeueAfter-driven — which is also the
-reference's steady state (its Create-only event filter makes creation a
-one-shot trigger; everything else is the timer,
-variantautoscaling_controller.go:473-486).
+Note on events: ``watch_create`` streams ADDED events over bounded watch
+windows (``?watch=true&timeoutSeconds=N``), which ManagerRuntime uses
+for Create-event wakeups — the same Create-only event filter the
+reference applies (variantautoscaling_controller.go:473-486; everything
+else remains RequeueAfter-driven).
 """
 
 from __future__ import annotations
@@ -186,3 +186,39 @@ class HTTPKubeClient:
     def delete(self, cls: Type[T], name: str, namespace: str) -> None:
         resp = self._client.delete(self._path(cls, namespace, name))
         self._raise_for(resp, f"delete {cls.__name__} {namespace}/{name}")
+
+    # ------------------------------------------------------------------ watch
+    def watch_create(
+        self, cls: Type[T], namespace: Optional[str] = None, timeout_seconds: int = 30
+    ):
+        """Yield objects of ``cls`` created while one watch window is open.
+
+        One bounded window: the generator returns when the server closes
+        the stream after ``timeout_seconds`` — callers loop to keep
+        watching (ManagerRuntime does, re-checking its stop flag between
+        windows).  Only ADDED events are consumed, matching the
+        reference's Create-only event filter.
+        """
+        import httpx
+        import json
+
+        prefix, plural, _ = self._resource(cls)
+        if namespace is None:
+            path = f"/{prefix}/{plural}"
+        else:
+            path = f"/{prefix}/namespaces/{namespace}/{plural}"
+        with self._client.stream(
+            "GET",
+            path,
+            params={"watch": "true", "timeoutSeconds": timeout_seconds},
+            timeout=httpx.Timeout(10.0, read=timeout_seconds + 10.0),
+        ) as resp:
+            if resp.status_code >= 400:
+                resp.read()
+                self._raise_for(resp, f"watch {cls.__name__}")
+            for line in resp.iter_lines():
+                if not line:
+                    continue
+                event = json.loads(line)
+                if event.get("type") == "ADDED":
+                    yield cls.model_validate(event["object"])

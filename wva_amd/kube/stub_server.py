@@ -1,13 +1,26 @@
 """Stub Kubernetes API server: the REST surface the controller uses,
 backed by an InMemoryKubeClient — the envtest analog for exercising
 HTTPKubeClient (and, through it, the whole controller) over real HTTP.
+
+List routes accept ``?watch=true`` and then stream newline-delimited
+``{"type": "ADDED", "object": ...}`` events for objects created while
+the connection is open (one bounded watch window of ``timeoutSeconds``).
+Divergence from the real API server, documented on purpose: no initial
+ADDED replay of existing objects — the consumer is the reconciler's
+Create-only event filter, where replay on every reconnect would fire a
+spurious wakeup per window.  Only ADDED is emitted; the reference
+ignores update/delete events anyway
+(variantautoscaling_controller.go:473-486).
 """
 
+import json as _json
+import queue as _queue
+import time
 from typing import Dict, Optional, Type
 
 from ..api.v1alpha1.types import VariantAutoscaling
 from fastapi import FastAPI, Request
-from fastapi.responses import JSONResponse
+from fastapi.responses import JSONResponse, StreamingResponse
 
 from .client import InMemoryKubeClient
 from .errors import ConflictError, NotFoundError
@@ -47,6 +60,34 @@ def create_stub_api_server(store: Optional[InMemoryKubeClient] = None):
         except Exception as e:  # validation and the rest
             return JSONResponse({"message": str(e)}, status_code=422)
 
+    def watch_stream(cls, namespace: Optional[str], timeout_s: int) -> StreamingResponse:
+        q: "_queue.Queue" = _queue.Queue()
+
+        def hook(obj):
+            if isinstance(obj, cls) and (
+                namespace is None or obj.metadata.namespace == namespace
+            ):
+                q.put(obj)
+
+        store.on_create(hook)
+
+        def gen():
+            try:
+                deadline = time.monotonic() + timeout_s
+                while True:
+                    remaining = deadline - time.monotonic()
+                    if remaining <= 0:
+                        return
+                    try:
+                        obj = q.get(timeout=min(remaining, 0.25))
+                    except _queue.Empty:
+                        continue
+                    yield _json.dumps({"type": "ADDED", "object": dump(obj)}) + "\n"
+            finally:
+                store.remove_create_hook(hook)
+
+        return StreamingResponse(gen(), media_type="application/json")
+
     for prefix, cls in _ROUTES.items():
         plural = _PLURALS[cls]
 
@@ -54,11 +95,15 @@ def create_stub_api_server(store: Optional[InMemoryKubeClient] = None):
             base = f"/{prefix}/namespaces/{{namespace}}/{plural}"
 
             @app.get(f"/{prefix}/{plural}")
-            async def list_all():
+            async def list_all(watch: bool = False, timeoutSeconds: int = 30):
+                if watch:
+                    return watch_stream(cls, None, timeoutSeconds)
                 return {"items": [dump(o) for o in store.list(cls)]}
 
             @app.get(base)
-            async def list_ns(namespace: str):
+            async def list_ns(namespace: str, watch: bool = False, timeoutSeconds: int = 30):
+                if watch:
+                    return watch_stream(cls, namespace, timeoutSeconds)
                 return {"items": [dump(o) for o in store.list(cls, namespace)]}
 
             @app.get(base + "/{name}")
