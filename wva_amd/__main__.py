@@ -6,9 +6,12 @@ Counterpart of /root/reference/cmd/main.go: flag surface
 probes, the controller metrics endpoint, custom metric registration, and
 the manager start.  Differences, deliberate:
 
-- leader election uses an exclusive flock on a lock file (there is no
-  API-server lease outside a cluster); the lock id mirrors the
-  reference's ``72dd1cf1.llm-d.ai``;
+- leader election: with ``--kube-backend in-cluster`` it is the
+  reference's protocol — a coordination/v1 Lease named
+  ``72dd1cf1.llm-d.ai`` renewed by the active manager
+  (wva_amd/controller/leader.py; lost leadership terminates the
+  process); the memory backend falls back to an exclusive flock on a
+  lock file with the same id (no API server to hold a lease);
 - HTTP/2 stays disabled by default (same CVE rationale,
   cmd/main.go:107-120) — uvicorn serves HTTP/1.1;
 - the Kubernetes backend is pluggable: ``--kube-backend memory`` runs
@@ -118,9 +121,6 @@ def main(argv=None) -> int:
     if not args.enable_http2:
         log.info("disabling http/2")
 
-    if args.leader_elect:
-        acquire_leader_lock(args.leader_lock_path)
-
     registry = CollectorRegistry()
     ctrl_metrics.init_metrics(registry)
 
@@ -130,6 +130,29 @@ def main(argv=None) -> int:
         client = HTTPKubeClient()  # service-account config
     else:
         client = InMemoryKubeClient()
+
+    elector = None
+    if args.leader_elect:
+        if args.kube_backend == "in-cluster":
+            # the reference's protocol: a coordination/v1 Lease named
+            # 72dd1cf1.llm-d.ai, renewed by the active manager
+            from .controller.constants import CONTROLLER_NAMESPACE
+            from .controller.leader import LeaseElector
+
+            elector = LeaseElector(
+                client,
+                namespace=os.environ.get("POD_NAMESPACE", CONTROLLER_NAMESPACE),
+            )
+
+            def _lost() -> None:
+                log.error("leader lease lost; terminating")
+                os._exit(1)
+
+            elector.on_lost = _lost
+            elector.acquire()
+        else:
+            # single-node dev backend: exclusive flock stands in
+            acquire_leader_lock(args.leader_lock_path)
 
     ready = {"ok": False}
     server = serve_http(args, registry, lambda: ready["ok"])
@@ -148,6 +171,8 @@ def main(argv=None) -> int:
         pass
     finally:
         runtime.stop()
+        if elector is not None:
+            elector.release()
         server.should_exit = True
     return 0
 

@@ -43,3 +43,6 @@ LABEL_VARIANT_NAME = "variant_name"
 LABEL_DIRECTION = "direction"
 LABEL_REASON = "reason"
 LABEL_ACCELERATOR_TYPE = "accelerator_type"
+
+# controller ConfigMap coordinates (shared by reconciler and leader election)
+CONTROLLER_NAMESPACE = "workload-variant-autoscaler-system"

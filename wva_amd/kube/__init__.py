@@ -14,7 +14,7 @@ ownerReference garbage collection, resourceVersion bumping) behind a small
 """
 
 from .errors import ConflictError, ForbiddenError, InvalidError, KubeError, NotFoundError
-from .objects import ConfigMap, Deployment, DeploymentSpec, DeploymentStatus
+from .objects import ConfigMap, Deployment, DeploymentSpec, DeploymentStatus, Lease, LeaseSpec
 from .client import InMemoryKubeClient, KubeClient
 from .http_client import HTTPKubeClient
 
@@ -26,6 +26,8 @@ __all__ = [
     "ConflictError",
     "ConfigMap",
     "Deployment",
+    "Lease",
+    "LeaseSpec",
     "DeploymentSpec",
     "DeploymentStatus",
     "KubeClient",

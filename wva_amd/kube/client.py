@@ -18,7 +18,7 @@ from typing import Callable, Dict, List, Optional, Protocol, Tuple, Type, TypeVa
 
 from ..api.v1alpha1.types import VariantAutoscaling
 from .errors import ConflictError, NotFoundError
-from .objects import ConfigMap, Deployment
+from .objects import ConfigMap, Deployment, Lease
 
 T = TypeVar("T")
 
@@ -26,6 +26,7 @@ _KINDS = {
     VariantAutoscaling: "VariantAutoscaling",
     ConfigMap: "ConfigMap",
     Deployment: "Deployment",
+    Lease: "Lease",
 }
 
 
@@ -98,6 +99,11 @@ class InMemoryKubeClient:
     def create(self, obj: T) -> T:
         with self._lock:
             key = self._key(obj)
+            if key in self._store:
+                # AlreadyExists is a Conflict in this package's error
+                # taxonomy (the API server's 409) — leader election's
+                # create race depends on it
+                raise ConflictError(f"{key} already exists")
             stored = copy.deepcopy(obj)
             stored.metadata.resource_version = next(self._rv)
             stored.metadata.uid = stored.metadata.uid or f"uid-{next(self._uid)}"

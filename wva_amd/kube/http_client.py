@@ -31,7 +31,7 @@ from typing import List, Optional, Type, TypeVar
 
 from ..api.v1alpha1.types import VariantAutoscaling
 from .errors import ConflictError, ForbiddenError, InvalidError, KubeError, NotFoundError
-from .objects import ConfigMap, Deployment
+from .objects import ConfigMap, Deployment, Lease
 
 T = TypeVar("T")
 
@@ -41,6 +41,7 @@ _RESOURCES = {
     VariantAutoscaling: ("apis/llmd.ai/v1alpha1", "variantautoscalings", True),
     ConfigMap: ("api/v1", "configmaps", False),
     Deployment: ("apis/apps/v1", "deployments", True),
+    Lease: ("apis/coordination.k8s.io/v1", "leases", True),
 }
 
 

@@ -1,7 +1,9 @@
-"""Typed Kubernetes objects the controller touches: ConfigMap, Deployment."""
+"""Typed Kubernetes objects the controller touches: ConfigMap,
+Deployment, and the coordination.k8s.io Lease used for leader election."""
 
 from __future__ import annotations
 
+import datetime
 from typing import Dict, Optional
 
 from pydantic import BaseModel, ConfigDict, Field
@@ -35,6 +37,33 @@ class DeploymentSpec(_Base):
 class DeploymentStatus(_Base):
     replicas: int = 0
     ready_replicas: int = Field(alias="readyReplicas", default=0)
+
+
+class LeaseSpec(_Base):
+    holder_identity: Optional[str] = Field(alias="holderIdentity", default=None)
+    lease_duration_seconds: Optional[int] = Field(alias="leaseDurationSeconds", default=None)
+    acquire_time: Optional[datetime.datetime] = Field(alias="acquireTime", default=None)
+    renew_time: Optional[datetime.datetime] = Field(alias="renewTime", default=None)
+    lease_transitions: int = Field(alias="leaseTransitions", default=0)
+
+
+class Lease(_Base):
+    """coordination.k8s.io/v1 Lease — leader-election primitive (the
+    reference's manager elects through the same object via
+    controller-runtime, LeaderElectionID 72dd1cf1.llm-d.ai)."""
+
+    api_version: str = Field(alias="apiVersion", default="coordination.k8s.io/v1")
+    kind: str = "Lease"
+    metadata: ObjectMeta = Field(default_factory=ObjectMeta)
+    spec: LeaseSpec = Field(default_factory=LeaseSpec)
+
+    @property
+    def name(self) -> str:
+        return self.metadata.name
+
+    @property
+    def namespace(self) -> str:
+        return self.metadata.namespace
 
 
 class Deployment(_Base):

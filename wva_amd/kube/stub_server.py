@@ -24,17 +24,19 @@ from fastapi.responses import JSONResponse, StreamingResponse
 
 from .client import InMemoryKubeClient
 from .errors import ConflictError, NotFoundError
-from .objects import ConfigMap, Deployment
+from .objects import ConfigMap, Deployment, Lease
 
 _ROUTES: Dict[str, Type] = {
     "apis/llmd.ai/v1alpha1": VariantAutoscaling,
     "api/v1": ConfigMap,
     "apis/apps/v1": Deployment,
+    "apis/coordination.k8s.io/v1": Lease,
 }
 _PLURALS = {
     VariantAutoscaling: "variantautoscalings",
     ConfigMap: "configmaps",
     Deployment: "deployments",
+    Lease: "leases",
 }
 
 
