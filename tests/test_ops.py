@@ -162,6 +162,23 @@ class TestGPU:
         assert_results_close(gpu, cpu, rtol=1e-4)
         assert gpu[:, R_FEASIBLE].sum() > 50
 
+    def test_hipgraph_replay_matches_plain(self, monkeypatch):
+        # same-shape solves replay a captured graph; fresh data must flow
+        # through the static buffers on every replay (stale-buffer check)
+        import wva_amd.ops.batched as B
+
+        monkeypatch.setattr(B, "_graph_cache", {})
+        monkeypatch.setattr(B, "_graph_disabled", False)
+        monkeypatch.setenv("WVA_GPU_GRAPH", "1")
+        for seed in (41, 42, 43):  # capture once, replay twice
+            problems = random_problems(96, seed=seed)
+            got = solve_problems(problems, device="cuda")
+            monkeypatch.setenv("WVA_GPU_GRAPH", "0")
+            plain = solve_problems(problems, device="cuda")
+            monkeypatch.setenv("WVA_GPU_GRAPH", "1")
+            np.testing.assert_array_equal(got, plain)
+        assert not B._graph_disabled  # capture really worked, no fallback
+
     def test_large_batch_limit_falls_back(self):
         problems = random_problems(8, seed=15)
         problems[0, 6] = 1024.0  # beyond the LDS-resident limit

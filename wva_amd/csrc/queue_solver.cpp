@@ -146,7 +146,7 @@ extern "C" void wva_launch_solve(const double *prob, double *out, int n_problems
                                  int max_k, void *stream);
 #endif
 
-static torch::Tensor solve_allocations(torch::Tensor problems) {
+static torch::Tensor solve_allocations(torch::Tensor problems, int64_t max_k_hint = -1) {
   TORCH_CHECK(problems.dim() == 2 && problems.size(1) == wva::PROBLEM_FIELDS,
               "problems must be [B, ", (int)wva::PROBLEM_FIELDS, "]");
   TORCH_CHECK(problems.scalar_type() == torch::kFloat64, "problems must be float64");
@@ -157,8 +157,16 @@ static torch::Tensor solve_allocations(torch::Tensor problems) {
 
   if (problems.is_cuda()) {
 #ifdef WVA_WITH_HIP
-    const double max_batch = problems.select(1, wva::P_MAX_BATCH).max().item<double>();
-    const int max_k = (int)max_batch * (1 + wva::kMaxQueueToBatchRatio);
+    int max_k;
+    if (max_k_hint > 0) {
+      // caller-supplied chain length (hipGraph capture cannot tolerate
+      // the .item() device sync below; wva_amd/ops/batched.py computes
+      // it host-side from the numpy batch before upload)
+      max_k = (int)max_k_hint;
+    } else {
+      const double max_batch = problems.select(1, wva::P_MAX_BATCH).max().item<double>();
+      max_k = (int)max_batch * (1 + wva::kMaxQueueToBatchRatio);
+    }
     TORCH_CHECK((max_k + 288) * 8 <= 64 * 1024,
                 "max_batch too large for the LDS-resident GPU path (limit ~700); "
                 "use the CPU path for these problems");
@@ -193,7 +201,8 @@ static torch::Tensor solve_allocations(torch::Tensor problems) {
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "wva_amd native batched queue solver (CPU + gfx950 HIP)";
   m.def("solve_allocations", &solve_allocations,
-        "Batched state-dependent M/M/1/K allocation sizing");
+        "Batched state-dependent M/M/1/K allocation sizing",
+        pybind11::arg("problems"), pybind11::arg("max_k_hint") = -1);
   m.attr("PROBLEM_FIELDS") = (int)wva::PROBLEM_FIELDS;
   m.attr("RESULT_FIELDS") = (int)wva::RESULT_FIELDS;
 #ifdef WVA_WITH_HIP

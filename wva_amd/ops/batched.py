@@ -45,6 +45,58 @@ RESULT_FIELDS = 6
 # the GPU path keeps the cumulative table in LDS: N <= ~700
 GPU_MAX_BATCH_LIMIT = 700
 
+# hipGraph dispatch cache: repeated fleet solves of the same shape replay
+# a captured (H2D -> kernel -> D2H-ready) graph instead of re-launching.
+# Keyed by (B, max_k); replays copy fresh problem data into the captured
+# input buffer.  WVA_GPU_GRAPH=0 disables; any capture failure falls back
+# to plain dispatch permanently (logged once).
+_graph_cache: dict = {}
+_graph_lock = None
+_graph_disabled = False
+
+
+def _graph_solve(native, problems_t, max_k):
+    """Solve via a cached hipGraph; None -> caller uses plain dispatch."""
+    global _graph_lock, _graph_disabled
+    import os
+    import threading
+
+    if _graph_disabled or os.environ.get("WVA_GPU_GRAPH", "1") == "0":
+        return None
+    import torch
+
+    if _graph_lock is None:
+        _graph_lock = threading.Lock()
+    key = (problems_t.shape[0], max_k)
+    with _graph_lock:
+        entry = _graph_cache.get(key)
+        try:
+            if entry is None:
+                static_in = problems_t.clone()
+                # warm up on a side stream (capture requires a clean stream)
+                side = torch.cuda.Stream()
+                side.wait_stream(torch.cuda.current_stream())
+                with torch.cuda.stream(side):
+                    native.solve_allocations(static_in, max_k)
+                torch.cuda.current_stream().wait_stream(side)
+                graph = torch.cuda.CUDAGraph()
+                with torch.cuda.graph(graph):
+                    static_out = native.solve_allocations(static_in, max_k)
+                _graph_cache[key] = (graph, static_in, static_out)
+                return static_out.cpu().numpy()
+            graph, static_in, static_out = entry
+            static_in.copy_(problems_t)
+            graph.replay()
+            return static_out.cpu().numpy()
+        except Exception as e:  # pragma: no cover - depends on ROCm graph support
+            _graph_disabled = True
+            import logging
+
+            logging.getLogger("wva").warning(
+                "hipGraph dispatch unavailable; using plain launches: %s", e
+            )
+            return None
+
 
 def _solve_problems_python(problems: np.ndarray) -> np.ndarray:
     """Reference-semantics scalar fallback via the Python analyzer."""
@@ -109,16 +161,22 @@ def solve_problems(problems: np.ndarray, device: Optional[str] = None) -> np.nda
         native = get_native()  # raises loudly if missing on a GPU box
         import torch
 
-        if problems[:, P_MAX_BATCH].max() > GPU_MAX_BATCH_LIMIT:
+        max_batch = problems[:, P_MAX_BATCH].max()
+        if max_batch > GPU_MAX_BATCH_LIMIT:
             # LDS-resident limit: solve oversized problems on CPU
             big = problems[:, P_MAX_BATCH] > GPU_MAX_BATCH_LIMIT
             out = np.empty((problems.shape[0], RESULT_FIELDS), dtype=np.float64)
             out[~big] = solve_problems(problems[~big], device=device)
             out[big] = solve_problems(problems[big], device="cpu")
             return out
+        # chain length computed host-side (also feeds hipGraph capture,
+        # where the binding's device-sync fallback is illegal)
+        max_k = int(max_batch) * (1 + MAX_QUEUE_TO_BATCH_RATIO)
         t = torch.from_numpy(problems).to(device)
-        res = native.solve_allocations(t)
-        return res.cpu().numpy()
+        res = _graph_solve(native, t, max_k)
+        if res is not None:
+            return res
+        return native.solve_allocations(t, max_k).cpu().numpy()
 
     if native_available():
         import torch
