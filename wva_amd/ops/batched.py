@@ -72,6 +72,10 @@ def _graph_solve(native, problems_t, max_k):
         entry = _graph_cache.get(key)
         try:
             if entry is None:
+                if len(_graph_cache) >= 16:
+                    # bound the private graph memory pools when fleet
+                    # shapes churn; oldest capture goes first
+                    _graph_cache.pop(next(iter(_graph_cache)))
                 static_in = problems_t.clone()
                 # warm up on a side stream (capture requires a clean stream)
                 side = torch.cuda.Stream()
