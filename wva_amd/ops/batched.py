@@ -46,10 +46,13 @@ RESULT_FIELDS = 6
 GPU_MAX_BATCH_LIMIT = 700
 
 # hipGraph dispatch cache: repeated fleet solves of the same shape replay
-# a captured (H2D -> kernel -> D2H-ready) graph instead of re-launching.
-# Keyed by (B, max_k); replays copy fresh problem data into the captured
-# input buffer.  WVA_GPU_GRAPH=0 disables; any capture failure falls back
-# to plain dispatch permanently (logged once).
+# a captured graph instead of re-launching.  Keyed by (B, max_k); replays
+# copy fresh problem data into the captured input buffer.  Opt-in via
+# WVA_GPU_GRAPH=1: the fleet solve is ONE kernel, so launch overhead is
+# ~1% of the dispatch and replay measured no win (0.96 vs 0.94 ms
+# headline cycle) — the path exists for launch-bound configurations
+# (many small heterogeneous dispatches), not the default.  Any capture
+# failure falls back to plain dispatch permanently (logged once).
 _graph_cache: dict = {}
 _graph_lock = None
 _graph_disabled = False
@@ -61,7 +64,7 @@ def _graph_solve(native, problems_t, max_k):
     import os
     import threading
 
-    if _graph_disabled or os.environ.get("WVA_GPU_GRAPH", "1") == "0":
+    if _graph_disabled or os.environ.get("WVA_GPU_GRAPH", "0") != "1":
         return None
     import torch
 
@@ -87,6 +90,9 @@ def _graph_solve(native, problems_t, max_k):
                 with torch.cuda.graph(graph):
                     static_out = native.solve_allocations(static_in, max_k)
                 _graph_cache[key] = (graph, static_in, static_out)
+                # capture only RECORDS the work — replay to actually run
+                # it for this call's data
+                graph.replay()
                 return static_out.cpu().numpy()
             graph, static_in, static_out = entry
             static_in.copy_(problems_t)
