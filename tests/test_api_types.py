@@ -104,3 +104,17 @@ class TestDefaults:
         assert v1alpha1.GROUP == "llmd.ai"
         assert v1alpha1.VERSION == "v1alpha1"
         assert v1alpha1.SHORT_NAME == "va"
+
+
+class TestCRDSchemaDoc:
+    def test_generated_doc_in_sync(self):
+        """docs/user-guide/crd-schema.md must match the pydantic types
+        (regenerate with `python hack/gen_crd_docs.py --write`)."""
+        import sys
+        from pathlib import Path
+
+        root = Path(__file__).resolve().parent.parent
+        sys.path.insert(0, str(root / "hack"))
+        import gen_crd_docs
+
+        assert gen_crd_docs.DOC_PATH.read_text() == gen_crd_docs.render()
