@@ -212,3 +212,54 @@ class TestEndToEnd:
                 # (metrics.go:118-124 parity) during the peak step
             finally:
                 scraper.stop()
+
+
+class TestShareGPTTrace:
+    def test_sharegpt_scaleup(self, registry):
+        """ShareGPT-like trace scale-up (the reference's OpenShift
+        hardware e2e, test/e2e-openshift/sharegpt_scaleup_test.go, run
+        in-process): heavy-tailed lognormal output lengths + variable
+        prompts ramping 1 -> 5 req/s must still produce a clean
+        scale-up with healthy conditions and a sane measured token mix."""
+        settings = EmulatorSettings(
+            model=MODEL,
+            decode_alpha=12.0,
+            decode_beta=6.0,
+            prefill_gamma=4.0,
+            prefill_delta=0.01,
+            avg_generated_len=25,
+            tokens_distribution="sharegpt",  # heavy tail, CV ~0.85
+            max_batch_size=16,
+            realtime=True,
+        )
+        cluster = make_cluster(opt_interval="1s")
+        make_deployment(cluster, replicas=1)
+        make_va(cluster, max_batch=16, alpha="12.0", beta="6.0", gamma="4.0", delta="0.01")
+
+        store = TimeSeriesStore()
+        scraper = Scraper(store)
+        prom = PromlibAPI(store)
+
+        with EmulatorProcess(settings) as emu:
+            scraper.add_target(f"{emu.base_url}/metrics", extra_labels={"namespace": "default"})
+            scraper.start(interval=0.5)
+            try:
+                rec = VariantAutoscalingReconciler(cluster, prom)
+                # ramp: light, then heavy with varied prompt lengths
+                drive_load(emu.base_url, rate_rps=1.0, duration_s=4.0, prompt_words=16)
+                rec.reconcile()
+                baseline = desired_replicas(cluster)
+                drive_load(emu.base_url, rate_rps=5.0, duration_s=8.0, prompt_words=48)
+                rec.reconcile()
+                peak = desired_replicas(cluster)
+                assert peak > baseline  # the trace forced a scale-up
+                assert simulate_hpa(cluster, registry) == peak
+                va = cluster.get(v1alpha1.VariantAutoscaling, "vllm-llama", "default")
+                assert v1alpha1.is_condition_true(va, v1alpha1.TYPE_METRICS_AVAILABLE)
+                assert v1alpha1.is_condition_true(va, v1alpha1.TYPE_OPTIMIZATION_READY)
+                # measured token mix reflects the trace, not the defaults:
+                # avg output must sit in the lognormal's plausible band
+                out_tokens = float(va.status.current_alloc.load.avg_output_tokens)
+                assert 5.0 <= out_tokens <= 200.0
+            finally:
+                scraper.stop()

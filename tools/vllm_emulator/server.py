@@ -9,6 +9,7 @@ style so tests can run several instances in-process via ASGI transport.
 from __future__ import annotations
 
 import asyncio
+import math
 import random
 import time
 import uuid
@@ -53,6 +54,13 @@ class OutputLengthSampler:
             return random.randint(0, 2 * self.avg)
         if self.distribution == "uniform-narrow":
             return random.randint(self.avg // 2, 3 * self.avg // 2)
+        if self.distribution == "sharegpt":
+            # heavy-tailed ShareGPT-like lengths: lognormal with CV~0.95
+            # (sigma 0.8), mean pinned to avg, capped at 4x to keep a
+            # single request from starving the batch
+            sigma = 0.8
+            mu = math.log(self.avg) - sigma * sigma / 2.0
+            return max(1, min(int(random.lognormvariate(mu, sigma)), 4 * self.avg))
         return self.avg  # deterministic
 
 
