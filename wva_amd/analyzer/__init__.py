@@ -52,4 +52,8 @@ __all__ = [
     "TargetPerf",
     "TargetRate",
     "effective_concurrency",
+    "MG1Corrector",
+    "MG1Metrics",
+    "pollaczek_khinchine_wait",
+    "service_scv_from_tokens",
 ]

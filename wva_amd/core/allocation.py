@@ -35,7 +35,6 @@ from ..config import (
     ACCEL_PENALTY_FACTOR,
     MAX_QUEUE_TO_BATCH_RATIO,
     AllocationData,
-    ServerLoadSpec,
 )
 
 if TYPE_CHECKING:  # pragma: no cover

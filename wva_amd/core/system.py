@@ -13,7 +13,6 @@ from typing import Dict, Optional
 from ..config import (
     AcceleratorCount,
     AcceleratorSpec,
-    AllocationData,
     AllocationSolution,
     ModelAcceleratorPerfData,
     OptimizerSpec,
