@@ -248,3 +248,57 @@ def test_webhook_cert_flags_accepted():
     args = parse_args(["--webhook-cert-path", "/certs"])
     assert args.webhook_cert_path == "/certs"
     assert args.webhook_cert_name == "tls.crt"
+
+
+class TestHelmChart:
+    CHART = DEPLOY.parent / "charts" / "workload-variant-autoscaler"
+
+    def test_chart_yaml(self):
+        chart = yaml.safe_load((self.CHART / "Chart.yaml").read_text())
+        assert chart["name"] == "workload-variant-autoscaler"
+        assert chart["apiVersion"] == "v2"
+
+    def test_crd_copy_matches_deploy(self):
+        # Helm installs crds/ once; the canonical schema lives in deploy/crd.
+        chart_crd = (self.CHART / "crds" / "llmd.ai_variantautoscalings.yaml").read_text()
+        deploy_crd = (DEPLOY / "crd" / "llmd.ai_variantautoscalings.yaml").read_text()
+        assert chart_crd == deploy_crd, "chart crds/ drifted from deploy/crd/"
+
+    def test_values_parse_and_accelerator_table(self):
+        values = yaml.safe_load((self.CHART / "values.yaml").read_text())
+        for name, payload in values["accelerators"].items():
+            entry = json.loads(payload)
+            assert "device" in entry and "cost" in entry, name
+        assert "MI355X" in values["accelerators"]
+        assert json.loads(values["accelerators"]["MI355X"])["memSize"] == "288"
+        for key, doc in values["serviceClasses"].items():
+            sc = yaml.safe_load(doc)
+            assert {"name", "priority", "data"} <= set(sc), key
+        dev = yaml.safe_load((self.CHART / "values-dev.yaml").read_text())
+        assert dev["va"]["enabled"] is True
+
+    def test_template_inventory(self):
+        import re
+
+        # every surface the reference chart covers must exist here
+        want = {
+            "deployment.yaml": "kind: Deployment",
+            "rbac.yaml": "kind: ClusterRole",
+            "leader-election-rbac.yaml": "kind: Role",
+            "rbac-aggregated.yaml": "aggregate-to-view",
+            "configmaps.yaml": "accelerator-unit-costs",
+            "metrics-service.yaml": "kind: Service",
+            "servicemonitor.yaml": "kind: ServiceMonitor",
+            "vllm-service.yaml": "kind: Service",
+            "vllm-servicemonitor.yaml": "kind: ServiceMonitor",
+            "hpa.yaml": "inferno_desired_replicas",
+            "variantautoscaling.yaml": "kind: VariantAutoscaling",
+            "prometheus-ca-configmap.yaml": "ca.crt",
+        }
+        for fname, marker in want.items():
+            text = (self.CHART / "templates" / fname).read_text()
+            assert marker in text, f"{fname} missing {marker!r}"
+            # helm delimiters balanced per template
+            assert text.count("{{") == text.count("}}"), fname
+        hpa = (self.CHART / "templates" / "hpa.yaml").read_text()
+        assert re.search(r"name: inferno_desired_replicas", hpa)
