@@ -273,3 +273,46 @@ class TestKVCacheCapacity:
             max_concurrent_tokens(288, profile, overhead_fraction=1.0)
         with pytest.raises(ValueError):
             max_batch_for_context(288, profile, 0)
+
+
+class TestAtTokensFallbackQuirk:
+    """De-facto contract from the reference (allocation.go:77-87 vs
+    internal/utils/utils.go:218-232): when the server carries no
+    maxBatchSize override, N falls back to max(perf.maxBatchSize *
+    atTokens / K, 1) — and the controller path never sets atTokens, so
+    the fallback degenerates to N=1, making the VA maxBatchSize
+    effectively mandatory."""
+
+    def _system(self, at_tokens):
+        from tests.fixtures import make_system, server_spec
+
+        system, _ = make_system(servers=[server_spec("s:ns", max_batch=0, out_tokens=128)])
+        perf = system.model("llama-8b").get_perf_data("MI355X")
+        perf.at_tokens = at_tokens
+        return system
+
+    def _explicit(self, max_batch):
+        from tests.fixtures import make_system, server_spec
+
+        system, _ = make_system(servers=[server_spec("s:ns", max_batch=max_batch, out_tokens=128)])
+        return system
+
+    def test_controller_path_degenerates_to_batch_1(self):
+        got = create_allocation(self._system(0), "s:ns", "MI355X")
+        want = create_allocation(self._explicit(1), "s:ns", "MI355X")
+        assert got is not None
+        assert got.num_replicas == want.num_replicas
+        assert got.itl == want.itl
+        assert got.ttft == want.ttft
+
+    def test_at_tokens_scales_derived_batch(self):
+        # N = perf.maxBatchSize * atTokens // K with atTokens=4*K (K=128)
+        system = self._system(4 * 128)
+        expect_n = system.model("llama-8b").get_perf_data("MI355X").max_batch_size * 4
+        got = create_allocation(system, "s:ns", "MI355X")
+        assert got.batch_size == expect_n
+        want = create_allocation(self._explicit(expect_n), "s:ns", "MI355X")
+        assert got is not None
+        assert got.num_replicas == want.num_replicas
+        assert got.itl == want.itl
+        assert got.max_arrv_rate_per_replica == want.max_arrv_rate_per_replica
