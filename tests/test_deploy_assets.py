@@ -302,3 +302,37 @@ class TestHelmChart:
             assert text.count("{{") == text.count("}}"), fname
         hpa = (self.CHART / "templates" / "hpa.yaml").read_text()
         assert re.search(r"name: inferno_desired_replicas", hpa)
+
+
+class TestKindEmulatorAssets:
+    KIND = DEPLOY / "kind-emulator"
+
+    def test_scripts_exist_and_are_executable(self):
+        import os
+
+        for script in ("setup.sh", "deploy-wva.sh", "teardown.sh"):
+            path = self.KIND / script
+            assert path.exists(), script
+            assert os.access(path, os.X_OK), f"{script} not executable"
+            assert path.read_text().startswith("#!/usr/bin/env bash"), script
+
+    def test_deploy_script_references_exist(self):
+        # every repo path the deploy script uses must exist
+        text = (self.KIND / "deploy-wva.sh").read_text()
+        repo = DEPLOY.parent
+        for rel in (
+            "tools/vllm_emulator/Dockerfile",
+            "hack/gen-tls-certs.sh",
+            "deploy/emulator/prometheus-tls-values.yaml",
+            "deploy/emulator/amd-gpu-node-labels.yaml",
+            "deploy/emulator/vllm-emulator.yaml",
+            "deploy/install.sh",
+            "deploy/samples/mi355x-variantautoscaling.yaml",
+        ):
+            assert (repo / rel).exists(), rel
+            assert rel.split("/")[-1] in text, f"{rel} not referenced"
+
+    def test_prometheus_tls_values_parse(self):
+        vals = yaml.safe_load((DEPLOY / "emulator" / "prometheus-tls-values.yaml").read_text())
+        tls = vals["prometheus"]["prometheusSpec"]["web"]["tlsConfig"]
+        assert tls["cert"]["secret"]["name"] == "prometheus-tls"
