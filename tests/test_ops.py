@@ -232,3 +232,79 @@ def test_concurrent_stream_submissions():
         outs = list(pool.map(run_on_stream, range(8)))
     for out in outs:
         np.testing.assert_array_equal(out, serial)
+
+
+class TestTorchFreeCPUBinding:
+    """_queue_native_cpu (pybind11/numpy/OpenMP): the slim controller
+    image's native sizing path; shares csrc/queue_host.h with the torch
+    binding so results must be bit-identical to it."""
+
+    @pytest.fixture(autouse=True)
+    def _need(self):
+        from wva_amd.ops import native_cpu_available
+
+        if not native_cpu_available():
+            pytest.skip("torch-free extension not built")
+
+    def test_identical_to_torch_binding(self):
+        if not native_available():
+            pytest.skip("torch extension not built")
+        import torch
+
+        from wva_amd.ops import get_native, get_native_cpu
+
+        problems = random_problems(128, seed=23)
+        a = get_native_cpu().solve_allocations(problems)
+        b = get_native().solve_allocations(torch.from_numpy(problems)).numpy()
+        np.testing.assert_array_equal(a, b)
+
+    def test_parity_vs_python_reference(self):
+        from wva_amd.ops import get_native_cpu
+
+        problems = random_problems(64, seed=29)
+        got = get_native_cpu().solve_allocations(problems)
+        want = _solve_problems_python(problems)
+        assert_results_close(got, want, rtol=1e-4)
+        assert got[:, R_FEASIBLE].sum() > 0
+
+    def test_shape_validation_and_empty(self):
+        from wva_amd.ops import get_native_cpu
+
+        cpu = get_native_cpu()
+        with pytest.raises(Exception):
+            cpu.solve_allocations(np.zeros((3, 5)))
+        assert cpu.solve_allocations(np.zeros((0, PROBLEM_FIELDS))).shape == (0, 6)
+
+    def test_used_when_torch_absent(self):
+        # the container scenario: torch unimportable -> solve_problems
+        # must route to the torch-free binding, not pure Python
+        import subprocess
+        import sys
+        from pathlib import Path
+
+        repo = Path(__file__).resolve().parent.parent
+        code = (
+            "import sys\n"
+            "class B:\n"
+            "    def find_module(self, n, p=None):\n"
+            "        if n == 'torch' or n.startswith('torch.'): return self\n"
+            "    def load_module(self, n): raise ImportError('blocked')\n"
+            "sys.meta_path.insert(0, B())\n"
+            "import numpy as np\n"
+            "import wva_amd.ops as ops\n"
+            "assert not ops.native_available()\n"
+            "assert ops.native_cpu_available()\n"
+            "from unittest import mock\n"
+            "p = np.zeros((2, ops.batched.PROBLEM_FIELDS)); p[:, 5] = 4; p[:, 6] = 2\n"
+            "p[:, 8] = 50.0; p[:, 10] = 1.0; p[:, 11] = 1\n"
+            "with mock.patch.object(ops.batched, '_solve_problems_python',\n"
+            "                       side_effect=AssertionError('python fallback used')):\n"
+            "    out = ops.solve_problems(p)\n"
+            "assert out.shape == (2, 6) and (out[:, 0] == 1.0).all()\n"
+            "print('torch-free routing OK')\n"
+        )
+        r = subprocess.run(
+            [sys.executable, "-c", code], cwd=repo, capture_output=True, text=True, timeout=120
+        )
+        assert r.returncode == 0, r.stdout + r.stderr
+        assert "torch-free routing OK" in r.stdout

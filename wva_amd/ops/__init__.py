@@ -21,9 +21,25 @@ try:  # pragma: no cover - import side effect
 except Exception as e:  # pragma: no cover
     _native_err = e
 
+# torch-free CPU binding (pybind11/numpy/OpenMP) — what the slim
+# controller image builds; same host solver (csrc/queue_host.h)
+_native_cpu = None
+try:  # pragma: no cover - import side effect
+    from wva_amd import _queue_native_cpu as _native_cpu  # type: ignore
+except Exception:  # pragma: no cover
+    pass
+
 
 def native_available() -> bool:
     return _native is not None
+
+
+def native_cpu_available() -> bool:
+    return _native_cpu is not None
+
+
+def get_native_cpu():
+    return _native_cpu
 
 
 def get_native():
@@ -43,4 +59,11 @@ def get_native():
 
 from .batched import BatchedAllocationSolver, solve_problems  # noqa: E402
 
-__all__ = ["native_available", "get_native", "BatchedAllocationSolver", "solve_problems"]
+__all__ = [
+    "native_available",
+    "native_cpu_available",
+    "get_native",
+    "get_native_cpu",
+    "BatchedAllocationSolver",
+    "solve_problems",
+]

@@ -193,6 +193,11 @@ def solve_problems(problems: np.ndarray, device: Optional[str] = None) -> np.nda
 
         native = get_native()
         return native.solve_allocations(torch.from_numpy(problems)).numpy()
+    from . import get_native_cpu, native_cpu_available
+
+    if native_cpu_available():
+        # torch-free binding (slim controller image): same host solver
+        return get_native_cpu().solve_allocations(problems)
     return _solve_problems_python(problems)
 
 
