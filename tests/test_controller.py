@@ -542,3 +542,34 @@ class TestOptimizationFailurePath:
         assert cond is not None and cond.status == "False"
         assert cond.reason == v1alpha1.REASON_OPTIMIZATION_FAILED
         assert va.status.desired_optimized_alloc.accelerator == ""
+
+
+class TestAnalyzerAutoWiring:
+    """WVA_BATCHED_ANALYZER / WVA_ANALYZER_DEVICE resolution: the deployed
+    controller should use the native batched sizing path whenever a
+    native binding is importable, without any flag."""
+
+    def test_auto_prefers_native_batched(self, cluster, prom, monkeypatch):
+        monkeypatch.delenv("WVA_BATCHED_ANALYZER", raising=False)
+        from wva_amd.ops import native_available, native_cpu_available
+
+        r = VariantAutoscalingReconciler(cluster, prom)
+        assert r.batched_analyzer == (native_available() or native_cpu_available())
+
+    def test_env_force_off(self, cluster, prom, monkeypatch):
+        monkeypatch.setenv("WVA_BATCHED_ANALYZER", "0")
+        r = VariantAutoscalingReconciler(cluster, prom)
+        assert r.batched_analyzer is False
+        assert r.analyzer_device is None
+
+    def test_env_force_on_with_device(self, cluster, prom, monkeypatch):
+        monkeypatch.setenv("WVA_BATCHED_ANALYZER", "1")
+        monkeypatch.setenv("WVA_ANALYZER_DEVICE", "cpu")
+        r = VariantAutoscalingReconciler(cluster, prom)
+        assert r.batched_analyzer is True
+        assert r.analyzer_device == "cpu"
+
+    def test_explicit_kwargs_win(self, cluster, prom, monkeypatch):
+        monkeypatch.setenv("WVA_BATCHED_ANALYZER", "1")
+        r = VariantAutoscalingReconciler(cluster, prom, batched_analyzer=False)
+        assert r.batched_analyzer is False

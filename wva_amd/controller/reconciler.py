@@ -23,6 +23,7 @@ native solver (one gfx950 dispatch per cycle) via ``batched_analyzer``.
 from __future__ import annotations
 
 import json
+import os
 import re
 import threading
 import time
@@ -93,11 +94,40 @@ class VariantAutoscalingReconciler:
         client: KubeClient,
         prom_api: Optional[PromAPI] = None,
         *,
-        batched_analyzer: bool = False,
+        batched_analyzer: Optional[bool] = None,
         analyzer_device: Optional[str] = None,
     ) -> None:
         self.client = client
         self.prom_api = prom_api
+        if batched_analyzer is None:
+            # WVA_BATCHED_ANALYZER: "1"/"0" force; unset = auto (batched
+            # when a native sizing binding is importable — the deployed
+            # controller then runs one native dispatch per cycle instead
+            # of the scalar Python loop; semantics are identical, see
+            # tests/test_controller.py::TestBatchedReconcile)
+            env = os.environ.get("WVA_BATCHED_ANALYZER", "")
+            if env in ("1", "true"):
+                batched_analyzer = True
+            elif env in ("0", "false"):
+                batched_analyzer = False
+            else:
+                from ..ops import native_available, native_cpu_available
+
+                batched_analyzer = native_available() or native_cpu_available()
+        if batched_analyzer and analyzer_device is None:
+            # WVA_ANALYZER_DEVICE: explicit "cuda"/"cpu"; unset = cuda
+            # when a GPU is visible and the torch extension is built
+            env_dev = os.environ.get("WVA_ANALYZER_DEVICE", "")
+            if env_dev:
+                analyzer_device = env_dev
+            else:
+                from ..ops import native_available
+
+                if native_available():
+                    import torch
+
+                    if torch.cuda.is_available():
+                        analyzer_device = "cuda"
         self.batched_analyzer = batched_analyzer
         self.analyzer_device = analyzer_device
         self.last_gpu_telemetry = {}
