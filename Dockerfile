@@ -18,13 +18,26 @@ RUN python setup.py build_ext --inplace
 
 FROM python:3.10-slim
 
-RUN pip install --no-cache-dir \
-        pydantic fastapi uvicorn httpx prometheus-client pyyaml numpy
+# libgomp1: _queue_native_cpu.so is built -fopenmp and links libgomp.so.1;
+# without it the import fails and sizing silently degrades to pure Python
+# (advisor r01 medium).
+RUN apt-get update && apt-get install -y --no-install-recommends libgomp1 \
+    && rm -rf /var/lib/apt/lists/*
+
+COPY requirements-image.txt /tmp/requirements-image.txt
+RUN pip install --no-cache-dir -r /tmp/requirements-image.txt
 
 RUN useradd --uid 65532 --no-create-home nonroot
 WORKDIR /app
 COPY --from=build /src/wva_amd/ wva_amd/
 
+# Image smoke check: the torch-free native sizing path must be importable
+# in this exact runtime (catches a missing libgomp1 / ABI drift at build
+# time rather than as a silent runtime fallback).
+RUN python -c "from wva_amd import ops; assert ops.native_cpu_available(), ops._native_cpu_err"
+
 USER 65532:65532
 EXPOSE 8443 8081
+# --kube-backend defaults to 'auto': in-cluster when KUBERNETES_SERVICE_HOST
+# is present (any pod), the in-memory dev backend otherwise.
 ENTRYPOINT ["python", "-m", "wva_amd"]

@@ -36,6 +36,28 @@ class TestCRD:
         cols = [(c["name"], c["jsonPath"]) for c in ver["additionalPrinterColumns"]]
         assert cols == v1alpha1.types.PRINT_COLUMNS
 
+    def test_required_lists_match_reference(self):
+        # The reference CRD rejects specs with empty perfParms and statuses
+        # missing currentAlloc fields; ours must too (advisor r01 low).
+        crd = load("crd/llmd.ai_variantautoscalings.yaml")
+        schema = crd["spec"]["versions"][0]["schema"]["openAPIV3Schema"]
+        spec = schema["properties"]["spec"]
+        acc_items = spec["properties"]["modelProfile"]["properties"][
+            "accelerators"]["items"]
+        assert acc_items["properties"]["perfParms"]["required"] == [
+            "decodeParms", "prefillParms"]
+        status = schema["properties"]["status"]["properties"]
+        assert sorted(status["currentAlloc"]["required"]) == [
+            "accelerator", "itlAverage", "load", "maxBatch",
+            "numReplicas", "ttftAverage", "variantCost"]
+        assert sorted(status["currentAlloc"]["properties"]["load"]["required"]) == [
+            "arrivalRate", "avgInputTokens", "avgOutputTokens"]
+        assert sorted(status["desiredOptimizedAlloc"]["required"]) == [
+            "accelerator", "numReplicas"]
+        assert status["actuation"]["required"] == ["applied"]
+        assert sorted(status["conditions"]["items"]["required"]) == [
+            "lastTransitionTime", "message", "reason", "status", "type"]
+
     def test_status_string_patterns(self):
         crd = load("crd/llmd.ai_variantautoscalings.yaml")
         status = crd["spec"]["versions"][0]["schema"]["openAPIV3Schema"]["properties"]["status"]
@@ -126,6 +148,41 @@ class TestMainEntry:
         assert args.leader_elect is False
         assert args.metrics_secure is True
         assert args.enable_http2 is False
+        assert args.kube_backend == "auto"  # in-cluster when in a pod
+
+    def test_deploy_args_pass_in_cluster_backend(self):
+        # A helm/kustomize install must never reconcile the in-memory fake
+        # (advisor r01 high): both deployment manifests pass the flag
+        # explicitly, and 'auto' also resolves to in-cluster inside a pod.
+        docs = load("controller.yaml")
+        dep = next(d for d in docs if d and d.get("kind") == "Deployment")
+        args = dep["spec"]["template"]["spec"]["containers"][0]["args"]
+        assert "--kube-backend=in-cluster" in args
+
+        chart_dep = (Path(__file__).resolve().parent.parent / "charts" /
+                     "workload-variant-autoscaler" / "templates" /
+                     "deployment.yaml").read_text()
+        assert "--kube-backend=in-cluster" in chart_dep
+
+    def test_auto_backend_resolves_in_cluster_inside_pod(self, monkeypatch):
+        from wva_amd.kube import InMemoryKubeClient
+        from wva_amd import __main__ as main_mod
+
+        made = {}
+
+        class FakeHTTP:
+            def __init__(self):
+                made["http"] = True
+                raise RuntimeError("stop before network")
+
+        monkeypatch.setenv("KUBERNETES_SERVICE_HOST", "10.0.0.1")
+        import wva_amd.kube as kube_mod
+        monkeypatch.setattr(kube_mod, "HTTPKubeClient", FakeHTTP)
+        try:
+            main_mod.main(["--max-cycles", "0"])
+        except RuntimeError:
+            pass
+        assert made.get("http") is True
 
     def test_leader_lock(self, tmp_path):
         from wva_amd.__main__ import acquire_leader_lock

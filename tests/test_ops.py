@@ -286,9 +286,12 @@ class TestTorchFreeCPUBinding:
         code = (
             "import sys\n"
             "class B:\n"
-            "    def find_module(self, n, p=None):\n"
-            "        if n == 'torch' or n.startswith('torch.'): return self\n"
-            "    def load_module(self, n): raise ImportError('blocked')\n"
+            "    # PEP 451 finder: find_spec raises, which aborts the import\n"
+            "    # on every supported Python (the legacy find_module hook is\n"
+            "    # gone in 3.12+)\n"
+            "    def find_spec(self, n, path=None, target=None):\n"
+            "        if n == 'torch' or n.startswith('torch.'):\n"
+            "            raise ImportError('blocked')\n"
             "sys.meta_path.insert(0, B())\n"
             "import numpy as np\n"
             "import wva_amd.ops as ops\n"

@@ -59,9 +59,12 @@ def parse_args(argv=None):
     ap.add_argument("--enable-http2", action="store_true", default=False)
     ap.add_argument(
         "--kube-backend",
-        choices=["memory", "in-cluster"],
-        default="memory",
-        help="'in-cluster' uses the mounted service account against the real API server",
+        choices=["memory", "in-cluster", "auto"],
+        default="auto",
+        help="'in-cluster' uses the mounted service account against the real "
+             "API server; 'auto' (default) resolves to in-cluster when "
+             "KUBERNETES_SERVICE_HOST is set (i.e. running in a pod) and to "
+             "the in-memory dev backend otherwise",
     )
     ap.add_argument("--max-cycles", type=int, default=None, help="exit after N reconcile cycles")
     return ap.parse_args(argv)
@@ -123,6 +126,15 @@ def main(argv=None) -> int:
 
     registry = CollectorRegistry()
     ctrl_metrics.init_metrics(registry)
+
+    if args.kube_backend == "auto":
+        # In a pod the kubelet exports KUBERNETES_SERVICE_HOST; a helm /
+        # kustomize install must never silently reconcile the in-memory
+        # fake (advisor r01: released image defaulted to 'memory').
+        args.kube_backend = (
+            "in-cluster" if os.environ.get("KUBERNETES_SERVICE_HOST") else "memory"
+        )
+        log.info("kube backend auto-resolved", backend=args.kube_backend)
 
     if args.kube_backend == "in-cluster":
         from .kube import HTTPKubeClient

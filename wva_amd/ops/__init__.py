@@ -24,10 +24,35 @@ except Exception as e:  # pragma: no cover
 # torch-free CPU binding (pybind11/numpy/OpenMP) — what the slim
 # controller image builds; same host solver (csrc/queue_host.h)
 _native_cpu = None
+_native_cpu_err: Exception | None = None
 try:  # pragma: no cover - import side effect
     from wva_amd import _queue_native_cpu as _native_cpu  # type: ignore
-except Exception:  # pragma: no cover
-    pass
+except Exception as e:  # pragma: no cover
+    _native_cpu_err = e
+
+_fallback_warned = False
+
+
+def warn_if_pure_python_fallback() -> None:
+    """One-time log when sizing runs pure-Python despite a broken extension.
+
+    A missing-by-design extension (dev checkout, never built) is silent; a
+    present-but-unloadable one (e.g. the libgomp1 case from the r01 image)
+    is surfaced so the degradation is diagnosable (advisor r01 low).
+    """
+    global _fallback_warned
+    if _fallback_warned:
+        return
+    _fallback_warned = True
+    err = _native_err if _native_err is not None else _native_cpu_err
+    if err is not None:
+        import logging
+
+        logging.getLogger("wva_amd.ops").warning(
+            "native sizing extension failed to import (%s: %s); "
+            "falling back to the pure-Python solver",
+            type(err).__name__, err,
+        )
 
 
 def native_available() -> bool:

@@ -202,6 +202,9 @@ def solve_problems(problems: np.ndarray, device: Optional[str] = None) -> np.nda
 
         native = get_native()
         return native.solve_allocations(torch.from_numpy(problems)).numpy()
+    from . import warn_if_pure_python_fallback
+
+    warn_if_pure_python_fallback()
     return _solve_problems_python(problems)
 
 
