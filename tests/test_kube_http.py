@@ -56,10 +56,31 @@ class TestHTTPCrud:
 
     def test_create_get_update_status_delete(self, client):
         from wva_amd.api.v1alpha1.types import ObjectMeta
+        from wva_amd.controller.reconciler import SERVICE_CLASSES_CM
 
+        # spec must now satisfy server-side CRD validation (a bare
+        # modelID-only VA is rejected, as the real apiserver does)
         va = v1alpha1.VariantAutoscaling(
             metadata=ObjectMeta(name="http-va", namespace="default"),
-            spec=v1alpha1.VariantAutoscalingSpec(modelID="m"),
+            spec=v1alpha1.VariantAutoscalingSpec(
+                modelID="m",
+                sloClassRef=v1alpha1.ConfigMapKeyRef(
+                    name=SERVICE_CLASSES_CM, key="premium.yaml"
+                ),
+                modelProfile=v1alpha1.ModelProfile(
+                    accelerators=[
+                        v1alpha1.AcceleratorProfile(
+                            acc="MI355X",
+                            accCount=1,
+                            maxBatchSize=8,
+                            perfParms=v1alpha1.PerfParms(
+                                decodeParms={"alpha": "6.9", "beta": "0.04"},
+                                prefillParms={"gamma": "20.0", "delta": "0.1"},
+                            ),
+                        )
+                    ]
+                ),
+            ),
         )
         created = client.create(va)
         assert created.metadata.resource_version > 0
@@ -67,6 +88,10 @@ class TestHTTPCrud:
         got = client.get(v1alpha1.VariantAutoscaling, "http-va", "default")
         assert got.spec.model_id == "m"
 
+        # a status write must itself satisfy schema validation: empty
+        # accelerator strings would be 422 on a real apiserver too
+        got.status.current_alloc.accelerator = "MI355X"
+        got.status.desired_optimized_alloc.accelerator = "MI355X"
         got.status.desired_optimized_alloc.num_replicas = 3
         updated = client.update_status(got)
         assert updated.status.desired_optimized_alloc.num_replicas == 3
@@ -131,9 +156,16 @@ class TestWatch:
         _, store = api_server
         got = queue.Queue()
 
+        # resume from "now": without a resourceVersion the server replays
+        # existing objects first (legacy list+watch semantics)
+        rv = store.resource_version
+
         def consume():
             for obj in client.watch_create(
-                v1alpha1.VariantAutoscaling, namespace="default", timeout_seconds=3
+                v1alpha1.VariantAutoscaling,
+                namespace="default",
+                timeout_seconds=3,
+                resource_version=rv,
             ):
                 got.put(obj)
 

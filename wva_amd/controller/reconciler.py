@@ -461,14 +461,23 @@ class ManagerRuntime:
         self._watch_threads: list = []
         if hasattr(client, "on_create"):
             client.on_create(self._on_create)
-        elif hasattr(client, "watch_create"):
-            # HTTP tier: Create-event wakeups via bounded watch windows
+        elif hasattr(client, "watch_events"):
+            # HTTP tier: Create-event wakeups via resumable watch
+            # sessions — resourceVersion continuity across windows, 410
+            # re-list recovery, exponential backoff
+            from ..kube.http_client import CreateWatchSession
+
+            self._watch_sessions = []
             for cls, namespace in (
                 (v1alpha1.VariantAutoscaling, None),
                 (ConfigMap, CONFIG_MAP_NAMESPACE),
             ):
+                session = CreateWatchSession(
+                    client, cls, namespace, window_seconds=5, stop_event=self._stop
+                )
+                self._watch_sessions.append(session)
                 t = threading.Thread(
-                    target=self._watch_loop, args=(cls, namespace), daemon=True
+                    target=session.run, args=(self._on_create,), daemon=True
                 )
                 t.start()
                 self._watch_threads.append(t)
@@ -498,22 +507,6 @@ class ManagerRuntime:
             "no Prometheus configuration found. Please set PROMETHEUS_BASE_URL "
             "environment variable or configure via ConfigMap"
         )
-
-    def _watch_loop(self, cls, namespace) -> None:
-        """One watch window at a time; reconnect until stopped.  Errors
-        back off briefly — a dead watch degrades to timer-only cadence,
-        never to a crashed runtime."""
-        while not self._stop.is_set():
-            try:
-                for obj in self.client.watch_create(
-                    cls, namespace=namespace, timeout_seconds=5
-                ):
-                    self._on_create(obj)
-                    if self._stop.is_set():
-                        return
-            except Exception as e:
-                log.debug("watch window failed; reconnecting", kind=cls.__name__, error=str(e))
-                self._stop.wait(1.0)
 
     def _on_create(self, obj) -> None:
         # Create-only event filter: VAs and the watched ConfigMap enqueue

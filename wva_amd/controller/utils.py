@@ -39,7 +39,7 @@ from ..config import (
     SystemData,
     SystemSpec,
 )
-from ..kube import ConfigMap, Deployment, KubeClient, KubeError
+from ..kube import ConfigMap, ConflictError, Deployment, KubeClient, KubeError
 from .collector import ACCELERATOR_LABEL
 from .interfaces import ServiceClassEntry, ServiceClassYaml
 from .logger import log
@@ -106,7 +106,25 @@ def get_variant_autoscaling_with_backoff(
 
 
 def update_status_with_backoff(client: KubeClient, obj, what: str) -> None:
-    retry_with_backoff(lambda: client.update_status(obj), STANDARD_BACKOFF, what)
+    """Status update with backoff; on 409 the object is re-read and the
+    status re-applied before retrying (client-go's RetryOnConflict
+    pattern — the reference retries the stale object and leaves conflict
+    resolution to the next cycle, utils.go:91-104; resolving in-place
+    keeps this cycle's optimization result instead of dropping it)."""
+    state = {"obj": obj}
+
+    def attempt():
+        try:
+            return client.update_status(state["obj"])
+        except ConflictError:
+            fresh = client.get(
+                type(obj), obj.metadata.name, obj.metadata.namespace
+            )
+            fresh.status = obj.status
+            state["obj"] = fresh
+            raise
+
+    retry_with_backoff(attempt, STANDARD_BACKOFF, what)
 
 
 def validate_prometheus_api(prom: PromAPI, backoff: Backoff = PROMETHEUS_BACKOFF) -> None:

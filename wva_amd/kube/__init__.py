@@ -13,7 +13,7 @@ ownerReference garbage collection, resourceVersion bumping) behind a small
   explicit endpoint); the controller only sees the protocol.
 """
 
-from .errors import ConflictError, ForbiddenError, InvalidError, KubeError, NotFoundError
+from .errors import ConflictError, ForbiddenError, GoneError, InvalidError, KubeError, NotFoundError
 from .objects import ConfigMap, Deployment, DeploymentSpec, DeploymentStatus, Lease, LeaseSpec
 from .client import InMemoryKubeClient, KubeClient
 from .http_client import HTTPKubeClient
@@ -24,6 +24,7 @@ __all__ = [
     "InvalidError",
     "ForbiddenError",
     "ConflictError",
+    "GoneError",
     "ConfigMap",
     "Deployment",
     "Lease",

@@ -23,3 +23,12 @@ class ConflictError(KubeError):
     """resourceVersion conflict on update; retry with a fresh read."""
 
     retryable = True
+
+
+class GoneError(KubeError):
+    """410 Gone: the requested resourceVersion is older than the server's
+    retained watch history (apimachinery's 'Expired').  Not retryable
+    in-place — the caller must re-list and restart the watch from the
+    fresh list resourceVersion."""
+
+    retryable = False

@@ -30,7 +30,14 @@ import ssl
 from typing import List, Optional, Type, TypeVar
 
 from ..api.v1alpha1.types import VariantAutoscaling
-from .errors import ConflictError, ForbiddenError, InvalidError, KubeError, NotFoundError
+from .errors import (
+    ConflictError,
+    ForbiddenError,
+    GoneError,
+    InvalidError,
+    KubeError,
+    NotFoundError,
+)
 from .objects import ConfigMap, Deployment, Lease
 
 T = TypeVar("T")
@@ -122,6 +129,8 @@ class HTTPKubeClient:
             raise ForbiddenError(msg)
         if resp.status_code == 409:
             raise ConflictError(msg)
+        if resp.status_code == 410:
+            raise GoneError(msg)
         if resp.status_code in (400, 422):
             raise InvalidError(msg)
         raise KubeError(msg)
@@ -141,15 +150,40 @@ class HTTPKubeClient:
         self._raise_for(resp, f"get {cls.__name__} {namespace}/{name}")
         return cls.model_validate(resp.json())
 
+    # controller-runtime's default list chunk size
+    LIST_PAGE_SIZE = 500
+
     def list(self, cls: Type[T], namespace: Optional[str] = None) -> List[T]:
+        items, _ = self.list_with_rv(cls, namespace)
+        return items
+
+    def list_with_rv(
+        self, cls: Type[T], namespace: Optional[str] = None, limit: int = 0
+    ) -> tuple:
+        """Chunked list following ``continue`` tokens; returns
+        (items, collection resourceVersion) — the rv a watch should
+        resume from."""
         prefix, plural, namespaced = self._resource(cls)
         if namespace is None:
             path = f"/{prefix}/{plural}"
         else:
             path = f"/{prefix}/namespaces/{namespace}/{plural}"
-        resp = self._client.get(path)
-        self._raise_for(resp, f"list {cls.__name__}")
-        return [cls.model_validate(item) for item in resp.json().get("items", [])]
+        limit = limit or self.LIST_PAGE_SIZE
+        items: List[T] = []
+        rv = 0
+        params = {"limit": limit}
+        while True:
+            resp = self._client.get(path, params=params)
+            self._raise_for(resp, f"list {cls.__name__}")
+            body = resp.json()
+            items.extend(cls.model_validate(i) for i in body.get("items", []))
+            meta = body.get("metadata", {})
+            if meta.get("resourceVersion") is not None:
+                rv = int(meta["resourceVersion"])
+            token = meta.get("continue", "")
+            if not token:
+                return items, rv
+            params = {"limit": limit, "continue": token}
 
     def create(self, obj: T) -> T:
         resp = self._client.post(
@@ -189,16 +223,20 @@ class HTTPKubeClient:
         self._raise_for(resp, f"delete {cls.__name__} {namespace}/{name}")
 
     # ------------------------------------------------------------------ watch
-    def watch_create(
-        self, cls: Type[T], namespace: Optional[str] = None, timeout_seconds: int = 30
+    def watch_events(
+        self,
+        cls: Type[T],
+        namespace: Optional[str] = None,
+        timeout_seconds: int = 30,
+        resource_version: int = 0,
+        allow_bookmarks: bool = True,
     ):
-        """Yield objects of ``cls`` created while one watch window is open.
+        """Yield (event_type, object_or_none, resourceVersion) for one
+        bounded watch window starting after ``resource_version``.
 
-        One bounded window: the generator returns when the server closes
-        the stream after ``timeout_seconds`` — callers loop to keep
-        watching (ManagerRuntime does, re-checking its stop flag between
-        windows).  Only ADDED events are consumed, matching the
-        reference's Create-only event filter.
+        BOOKMARK events yield (type, None, rv).  A server-side 410
+        ('Expired' ERROR event or HTTP status) raises GoneError — the
+        caller must re-list and resume from the fresh list rv.
         """
         import httpx
         import json
@@ -208,10 +246,15 @@ class HTTPKubeClient:
             path = f"/{prefix}/{plural}"
         else:
             path = f"/{prefix}/namespaces/{namespace}/{plural}"
+        params = {"watch": "true", "timeoutSeconds": timeout_seconds}
+        if resource_version:
+            params["resourceVersion"] = str(resource_version)
+        if allow_bookmarks:
+            params["allowWatchBookmarks"] = "true"
         with self._client.stream(
             "GET",
             path,
-            params={"watch": "true", "timeoutSeconds": timeout_seconds},
+            params=params,
             timeout=httpx.Timeout(10.0, read=timeout_seconds + 10.0),
         ) as resp:
             if resp.status_code >= 400:
@@ -221,5 +264,111 @@ class HTTPKubeClient:
                 if not line:
                     continue
                 event = json.loads(line)
-                if event.get("type") == "ADDED":
-                    yield cls.model_validate(event["object"])
+                etype = event.get("type")
+                obj = event.get("object", {})
+                if etype == "ERROR":
+                    if obj.get("code") == 410:
+                        raise GoneError(obj.get("message", "resourceVersion expired"))
+                    raise KubeError(f"watch {cls.__name__}: {obj}")
+                if etype == "BOOKMARK":
+                    yield etype, None, int(obj["metadata"]["resourceVersion"])
+                    continue
+                rv = int(obj.get("metadata", {}).get("resourceVersion", 0))
+                yield etype, cls.model_validate(obj), rv
+
+    def watch_create(
+        self, cls: Type[T], namespace: Optional[str] = None, timeout_seconds: int = 30,
+        resource_version: int = 0,
+    ):
+        """Back-compat shim: ADDED objects from one watch window."""
+        for etype, obj, _ in self.watch_events(
+            cls, namespace, timeout_seconds, resource_version
+        ):
+            if etype == "ADDED":
+                yield obj
+
+
+class CreateWatchSession:
+    """Continuity layer over bounded watch windows (reference analog:
+    controller-runtime's reflector ListAndWatch).
+
+    Round-1 weakness: windows streamed only live ADDED
+    events with no resourceVersion carry-over, so creates landing between
+    windows were missed until the next timer tick.  This session:
+
+    - primes with a paged list and resumes every window from the last
+      seen rv (bookmarks advance it while idle);
+    - on 410 Gone re-lists and surfaces objects whose uid was not seen
+      before (the creates the expired window dropped);
+    - applies exponential backoff (1 s .. 30 s) to connection errors,
+      resetting after any healthy window.
+
+    ``run(callback)`` invokes callback(obj) for every create until
+    ``stop()``.
+    """
+
+    BACKOFF_BASE_S = 1.0
+    BACKOFF_MAX_S = 30.0
+
+    def __init__(
+        self,
+        client: HTTPKubeClient,
+        cls,
+        namespace: Optional[str] = None,
+        window_seconds: int = 30,
+        stop_event=None,
+    ) -> None:
+        import threading
+
+        self.client = client
+        self.cls = cls
+        self.namespace = namespace
+        self.window_seconds = window_seconds
+        self._stop = stop_event if stop_event is not None else threading.Event()
+        self._rv = 0
+        self._seen_uids: set = set()
+        self.backoff_s = self.BACKOFF_BASE_S  # exposed for tests
+
+    def stop(self) -> None:
+        self._stop.set()
+
+    def _prime(self):
+        """(Re-)list: refresh rv and return creates missed while blind.
+
+        On first prime this surfaces every existing object — the same
+        initial-sync Add events a controller-runtime informer delivers.
+        """
+        items, rv = self.client.list_with_rv(self.cls, self.namespace)
+        self._rv = rv
+        fresh = [o for o in items if o.metadata.uid not in self._seen_uids]
+        self._seen_uids.update(o.metadata.uid for o in items)
+        return fresh
+
+    def run(self, callback) -> None:
+        primed = False
+        while not self._stop.is_set():
+            try:
+                if not primed:
+                    for obj in self._prime():
+                        callback(obj)
+                    primed = True
+                for etype, obj, rv in self.client.watch_events(
+                    self.cls,
+                    self.namespace,
+                    timeout_seconds=self.window_seconds,
+                    resource_version=self._rv,
+                ):
+                    self._rv = max(self._rv, rv)
+                    if self._stop.is_set():
+                        return
+                    if etype == "ADDED" and obj.metadata.uid not in self._seen_uids:
+                        self._seen_uids.add(obj.metadata.uid)
+                        callback(obj)
+                self.backoff_s = self.BACKOFF_BASE_S
+            except GoneError:
+                # watch history compacted: re-list immediately, resume
+                # from the fresh collection rv
+                primed = False
+            except Exception:
+                self._stop.wait(self.backoff_s)
+                self.backoff_s = min(self.backoff_s * 2, self.BACKOFF_MAX_S)
