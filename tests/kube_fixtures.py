@@ -90,6 +90,7 @@ def make_va(
     beta="0.042",
     gamma="20.0",
     delta="0.1",
+    acc_count=1,
 ):
     va = v1alpha1.VariantAutoscaling(
         metadata=ObjectMeta(
@@ -104,7 +105,7 @@ def make_va(
                 accelerators=[
                     v1alpha1.AcceleratorProfile(
                         acc=accelerator,
-                        accCount=1,
+                        accCount=acc_count,
                         maxBatchSize=max_batch,
                         perfParms=v1alpha1.PerfParms(
                             decodeParms={"alpha": alpha, "beta": beta},

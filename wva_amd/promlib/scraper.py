@@ -34,8 +34,13 @@ class Scraper:
         self._stop = threading.Event()
 
     def add_target(self, fetch: Union[Fetcher, str], extra_labels: Optional[Dict[str, str]] = None) -> None:
+        labels = dict(extra_labels or {})
         if isinstance(fetch, str):
             url = fetch
+            # Prometheus attaches a unique `instance` label per scrape
+            # target; without it two pods of one model collide into one
+            # series and sum(rate(...)) under-counts the fleet
+            labels.setdefault("instance", url.split("//")[-1].split("/")[0])
 
             def http_fetch() -> bytes:
                 import httpx
@@ -43,7 +48,9 @@ class Scraper:
                 return httpx.get(url, timeout=5.0).content
 
             fetch = http_fetch
-        self._targets.append(Target(fetch, extra_labels))
+        else:
+            labels.setdefault("instance", f"target-{len(self._targets)}")
+        self._targets.append(Target(fetch, labels))
 
     def scrape_once(self, ts: Optional[float] = None) -> None:
         ts = time.time() if ts is None else ts
