@@ -235,6 +235,8 @@ def main() -> None:
     ap.add_argument("--warmup", type=int, default=5)
     ap.add_argument("--variants-per-gpu", type=int, default=VARIANTS_PER_GPU)
     ap.add_argument("--backend", default="", help="override torch.distributed backend (default: nccl on GPU, gloo on CPU)")
+    ap.add_argument("--no-observed-slo", action="store_true",
+                    help="skip the untimed closed-loop observed-SLO measurement")
     args = ap.parse_args()
 
     world_size = int(os.environ.get("WORLD_SIZE", "1"))
@@ -312,6 +314,32 @@ def main() -> None:
     unlimited_ms_per_step = elapsed_unlimited / args.steps * 1000.0
     slo_attainment = 100.0 * slo_met / slo_total if slo_total else 0.0
 
+    # Observed SLO attainment (untimed, single-rank runs only): a short
+    # closed loop against the scalable emulator fleet — load -> scrape ->
+    # solve -> resize -> measured TTFT/ITL vs targets (VERDICT r01 #2:
+    # attainment measured from serving latency, not from the sizing
+    # model's own predictions).
+    observed = {}
+    if rank == 0 and world_size == 1 and not args.no_observed_slo:
+        sys.path.insert(0, str(Path(__file__).resolve().parent / "tools"))
+        sys.path.insert(0, str(Path(__file__).resolve().parent / "tests"))
+        try:
+            from soak import run_soak
+
+            soak_result = run_soak(
+                stages=(2.0, 4.0, 6.0, 4.0), stage_seconds=5.0,
+                max_replicas=4, quiet=True,
+            )
+            observed = {
+                "observed_slo_attainment_pct": soak_result[
+                    "observed_slo_attainment_pct"
+                ],
+                "mean_itl_drift_pct": soak_result["mean_itl_drift_pct"],
+            }
+        except Exception as e:  # pragma: no cover - env-dependent
+            observed = {"observed_slo_attainment_pct": None,
+                        "observed_slo_error": str(e)[:200]}
+
     if rank == 0:
         print(
             json.dumps(
@@ -330,6 +358,7 @@ def main() -> None:
                     "data": "synthetic",
                     "slo_attainment_pct": slo_attainment,
                     "unlimited_ms_per_step": unlimited_ms_per_step,
+                    **observed,
                     "config": {
                         "model": "3 models x 2 service classes (config #3)",
                         "variants": n_gpus * n_variants,

@@ -34,14 +34,20 @@ def check_contract(out, n_gpus):
 class TestBench:
     def test_single_process(self):
         proc = subprocess.run(
-            [sys.executable, "bench.py", "--steps", "2", "--warmup", "1", "--variants-per-gpu", "8"],
+            [sys.executable, "bench.py", "--steps", "2", "--warmup", "1",
+             "--variants-per-gpu", "8", "--no-observed-slo"],
             cwd=ROOT,
             capture_output=True,
             text=True,
             timeout=300,
         )
         assert proc.returncode == 0, proc.stderr
-        check_contract(last_json_line(proc.stdout), n_gpus=1)
+        out = last_json_line(proc.stdout)
+        check_contract(out, n_gpus=1)
+        # the headline is the capacity-constrained greedy solve; the
+        # easier unlimited argmin rides along as a secondary series
+        assert out["unlimited_ms_per_step"] > 0
+        assert "greedy-limited" in out["config"]["solver_mode"]
 
     def test_two_rank_gloo(self):
         env = dict(os.environ, MASTER_ADDR="127.0.0.1")

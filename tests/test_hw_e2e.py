@@ -56,8 +56,12 @@ def test_gpu_analyzer_scaleup_chain(registry, monkeypatch):
     scraper = Scraper(store)
     prom = PromlibAPI(store)
 
-    with EmulatorProcess(settings) as emu:
-        scraper.add_target(f"{emu.base_url}/metrics", extra_labels={"namespace": "default"})
+    from vllm_emulator.fleet import EmulatorFleet
+    from slo_observer import observe_latency
+
+    with EmulatorFleet(settings, max_replicas=4) as fleet:
+        for url in fleet.urls:
+            scraper.add_target(f"{url}/metrics", extra_labels={"namespace": "default"})
         scraper.start(interval=0.5)
         try:
             # the analyze phase runs on the GPU (fails loudly if the native
@@ -67,7 +71,10 @@ def test_gpu_analyzer_scaleup_chain(registry, monkeypatch):
             )
             time.sleep(2.0)
             rec.reconcile()
-            drive_load(emu.base_url, rate_rps=5.0, duration_s=6.0)
+            fleet.drive(5.0, 6.0, "default/llama-8b")
+            # under-provisioned: one instance serves 5 rps against a
+            # binding ITL SLO — the observed latency breaches the target
+            obs_before = observe_latency(prom, "default/llama-8b", "default")
             rec.reconcile()
             va = cluster.get(v1alpha1.VariantAutoscaling, "vllm-llama", "default")
             desired = va.status.desired_optimized_alloc.num_replicas
@@ -76,5 +83,20 @@ def test_gpu_analyzer_scaleup_chain(registry, monkeypatch):
             # external autoscaler follows the gauge; Deployment resizes
             assert simulate_hpa(cluster, registry) == desired
             assert cluster.get(Deployment, "vllm-llama", "default").spec.replicas == desired
+
+            # actuation closes the loop: after the fleet grows to the
+            # recommendation, OBSERVED serving latency returns under the
+            # SLO (the reference's hardware-e2e relation,
+            # sharegpt_scaleup_test.go:39-253) — measured, not predicted
+            fleet.scale(desired)
+            fleet.drive(5.0, 6.0, "default/llama-8b")
+            obs_after = observe_latency(prom, "default/llama-8b", "default")
+            target_itl = 24.0  # Premium slo-tpot (kube_fixtures.py)
+            assert obs_after.itl_ms <= target_itl * 1.1, (
+                f"scaled fleet still over ITL target: {obs_after.itl_ms:.1f}ms"
+            )
+            assert obs_after.itl_ms <= obs_before.itl_ms + 1.0, (
+                "scaling should not worsen observed ITL"
+            )
         finally:
             scraper.stop()
