@@ -55,7 +55,9 @@ def main():
             t = torch.from_numpy(problems).cuda()
             entry = {"batch": B, "N": n_hi, "K": 11 * n_hi}
             results = {}
-            for threads in (64, 128, 256, None):
+            for threads in (64, 128, 256, 384, 896, None):
+                if threads == 896 and 11 * n_hi + 896 + 64 > 8192:
+                    continue  # exceeds the 64 KiB LDS budget; launcher would substitute
                 if threads is None:
                     os.environ.pop("WVA_GPU_THREADS", None)  # launcher auto-select
                 else:
@@ -73,16 +75,17 @@ def main():
             # the rate column on feasible rows
             cpu = solve_problems(problems, device="cpu")
             base = results[64]
+            geoms = [g for g in (128, 256, 384, 896, None) if g in results]
             entry["geom_feas_agree"] = all(
                 (results[g][:, R_FEASIBLE] == base[:, R_FEASIBLE]).all()
-                for g in (128, 256, None)
+                for g in geoms
             )
             feas = base[:, R_FEASIBLE] == 1.0
             entry["max_rel_rate_diff"] = max(
                 float(np.abs(
                     (results[g][feas, 2] - base[feas, 2]) / np.maximum(base[feas, 2], 1e-300)
                 ).max()) if feas.any() else 0.0
-                for g in (128, 256, None)
+                for g in geoms
             )
             entry["cpu_feas_flips"] = int((base[:, R_FEASIBLE] != cpu[:, R_FEASIBLE]).sum())
             entry["feasible_frac"] = float(base[:, R_FEASIBLE].mean())

@@ -386,6 +386,253 @@ __device__ void solve_body_dual(const double *__restrict__ prob, double *__restr
   }
 }
 
+// ---------------------------------------------------------------------------
+// Speculative-tree multisection ("spec") kernel — round-2 lever
+// (profiles/r01_rocprof_pmc.txt: ~32% of wave time parked at barriers and
+// ~34% issue-stalled on the fp64 exp/log chain of ~200 SEQUENTIAL model
+// evaluations; the bisection itself is the critical path).
+//
+// Geometry: 2 search groups (TTFT, ITL) x SW waves, SW = 2^D - 1.  Each
+// round every group evaluates the ENTIRE depth-D midpoint tree of its
+// current bracket concurrently — candidate k (heap order) is the
+// midpoint reached after the D halvings encoded by k's bits, computed by
+// the same (lo+hi)/2 recurrence the sequential loop uses, so the
+// iterates are BITWISE identical to wg_binary_search and parity is
+// exact, not approximate.  The group then replays the sequential
+// decision walk on the cached y values, consuming D bisection levels per
+// round: D=2 (SW=3, 384 threads) turns ~100 serial evaluations per
+// target into ~50 rounds, with 6 resident waves per CU hiding the exp
+// chain latency the PMC profile flagged.
+//
+// Wave synchronization is group-local LDS mailboxes (publish y, fence,
+// bump flag; consumers spin with s_sleep) — NO workgroup barriers after
+// the scan, so the two searches never lock-step each other (the dual
+// kernel's exchange barrier is gone too).
+// ---------------------------------------------------------------------------
+
+__device__ inline void mb_publish(volatile double *slot, double y,
+                                  volatile double *flag, double round_id) {
+  *slot = y;
+  __threadfence_block();
+  *flag = round_id;
+}
+
+__device__ inline void mb_wait(volatile double *flags, int n, double round_id) {
+  for (int j = 0; j < n; ++j) {
+    while (flags[j] < round_id) __builtin_amdgcn_s_sleep(2);
+  }
+  __threadfence_block();
+}
+
+// Midpoint at heap node k (1-based: 1 = root midpoint, 2/3 = children,
+// 4..7 = grandchildren).  The bits of k below its leading 1 encode the
+// left(0)/right(1) path; each step repeats the sequential loop's exact
+// (lo+hi)/2 recurrence, so the value is BITWISE what the sequential
+// bisection would compute on that path.
+__device__ inline double heap_midpoint(double lo, double hi, int k) {
+  double x = 0.5 * (lo + hi);
+  const int depth_below = 31 - __clz(k);
+  for (int level = depth_below - 1; level >= 0; --level) {
+    if ((k >> level) & 1) {
+      lo = x;
+    } else {
+      hi = x;
+    }
+    x = 0.5 * (lo + hi);
+  }
+  return x;
+}
+
+// Speculative bisection for one search group.  All SW waves of the group
+// execute this with identical control flow; `sub` is the wave's index
+// within the group, `yb`/`fl` its SW-slot mailbox.  Returns the boundary
+// indicator (-1/0/+1) and writes lambda* like wg_binary_search.
+template <int D, typename F>
+__device__ int spec_binary_search(double x_min, double x_max, double y_target,
+                                  F eval, int sub, volatile double *yb,
+                                  volatile double *fl, double *x_star) {
+  constexpr int SW = (1 << D) - 1;
+  const int lane = threadIdx.x & 63;
+
+  // pre-phase: y(x_min) and y(x_max) evaluated concurrently (slots 0,1)
+  double mine = 0.0;
+  if (sub == 0) mine = eval(x_min);
+  if (sub == 1) mine = eval(x_max);
+  if (lane == 0 && sub < 2) mb_publish(&yb[sub], mine, &fl[sub], 1.0);
+  mb_wait(fl, 2, 1.0);
+  const double y0 = yb[0];
+  const double y1 = yb[1];
+
+  if (within_tolerance(y0, y_target, kTolerance)) {
+    *x_star = x_min;
+    return 0;
+  }
+  if (within_tolerance(y1, y_target, kTolerance)) {
+    *x_star = x_max;
+    return 0;
+  }
+  if (within_tolerance(y0, y1, kTolerance)) {
+    if (y_target > fmax(y0, y1)) {
+      *x_star = x_max;
+      return +1;
+    }
+    *x_star = x_min;
+    return -1;
+  }
+  const bool increasing = y0 < y1;
+  if ((increasing && y_target < y0) || (!increasing && y_target > y0)) {
+    *x_star = x_min;
+    return -1;
+  }
+  if ((increasing && y_target > y1) || (!increasing && y_target < y1)) {
+    *x_star = x_max;
+    return +1;
+  }
+
+  double lo = x_min, hi = x_max, xs = x_min;
+  double round_id = 2.0;
+  int iters = 0;
+  while (iters < kMaxIterations) {
+    // evaluate the whole depth-D midpoint tree of [lo, hi]
+    const double x_mine = heap_midpoint(lo, hi, sub + 1);
+    const double y_mine = eval(x_mine);
+    if (lane == 0) mb_publish(&yb[sub], y_mine, &fl[sub], round_id);
+    mb_wait(fl, SW, round_id);
+    round_id += 1.0;
+
+    // replay the sequential walk on the cached tree
+    int k = 1;
+    bool done = false;
+    for (int level = 0; level < D && iters < kMaxIterations; ++level) {
+      xs = 0.5 * (lo + hi);
+      const double ys = yb[k - 1];
+      ++iters;
+      if (within_tolerance(ys, y_target, kTolerance)) {
+        done = true;
+        break;
+      }
+      if ((increasing && y_target < ys) || (!increasing && y_target > ys)) {
+        hi = xs;
+        k = 2 * k;
+      } else {
+        lo = xs;
+        k = 2 * k + 1;
+      }
+    }
+    if (done) break;
+  }
+  *x_star = xs;
+  return 0;
+}
+
+// Mailbox LDS layout (doubles, after cum[max_k] + totals[THREADS]):
+//   [0..6]   group-0 y slots      [7..13]  group-0 flags
+//   [14..20] group-1 y slots      [21..27] group-1 flags
+//   [28..31] results: lam0, ind0, lam1, ind1
+//   [32..33] done flags per group
+template <int D>
+__device__ void solve_body_spec(const double *__restrict__ prob,
+                                double *__restrict__ out, int n_problems,
+                                int max_k) {
+  constexpr int SW = (1 << D) - 1;
+  constexpr int THREADS = 2 * SW * 64;
+  const int pid = blockIdx.x;
+  if (pid >= n_problems) return;
+  const double *pr = prob + (size_t)pid * PROBLEM_FIELDS;
+  double *res = out + (size_t)pid * RESULT_FIELDS;
+  const int tid = threadIdx.x;
+  const int wave = tid >> 6;
+  const int group = wave / SW;   // 0 = TTFT, 1 = ITL
+  const int sub = wave % SW;
+  const int lane = tid & 63;
+
+  const Parms p = load_parms(pr);
+  const int K = p.max_batch * (1 + kMaxQueueToBatchRatio);
+
+  extern __shared__ double smem[];
+  double *cum = smem;
+  double *totals = smem + max_k;
+  double *mail = totals + THREADS;
+  if (tid < 34) mail[tid] = 0.0;
+
+  build_cum<THREADS>(p, K, cum, totals);  // ends with __syncthreads()
+
+  const double lam_min = serv_rate(p, 1) * kEpsilon;  // req/ms
+  const double lam_max = serv_rate(p, p.max_batch) * (1.0 - kEpsilon);
+
+  WgEval<64> ev{p, cum, nullptr, K};  // lane-indexed, barrier-free
+
+  if (tid < RESULT_FIELDS) res[tid] = 0.0;
+
+  volatile double *yb = mail + group * 14;
+  volatile double *fl = yb + 7;
+  volatile double *ex = mail + 28;
+  volatile double *done = mail + 32;
+
+  double lam_t = lam_max;
+  int ind = 0;
+  const double target = (group == 0) ? pr[P_TARGET_TTFT] : pr[P_TARGET_ITL];
+  if (target > 0.0) {
+    if (group == 0) {
+      ind = spec_binary_search<D>(
+          lam_min, lam_max, target,
+          [&](double x) { return ev.eval_ttft(x); }, sub, yb, fl, &lam_t);
+    } else {
+      ind = spec_binary_search<D>(
+          lam_min, lam_max, target,
+          [&](double x) { return ev.eval_itl(x); }, sub, yb, fl, &lam_t);
+    }
+  }
+  if (sub == 0 && lane == 0) {
+    ex[group * 2 + 0] = lam_t;
+    ex[group * 2 + 1] = (double)ind;
+    __threadfence_block();
+    done[group] = 1.0;
+  }
+  if (wave != 0) return;  // group-0 wave 0 finishes the tail alone
+
+  mb_wait(done, 2, 1.0);
+  if (ex[1] < 0.0 || ex[3] < 0.0) return;  // a target below reachable range
+
+  double lam_tps = lam_max;
+  if (pr[P_TARGET_TPS] > 0.0) lam_tps = lam_max * (1.0 - kStabilityFraction);
+
+  const double lam = fmin(ex[0], fmin(ex[2], lam_tps));
+  Stats st = ev.eval(lam);
+  const double rate_star = st.throughput * 1000.0;  // req/s
+
+  const double total_rate = pr[P_TOTAL_RATE];
+  double n_rep = ceil(total_rate / rate_star);
+  if (n_rep < pr[P_MIN_REPLICAS]) n_rep = pr[P_MIN_REPLICAS];
+  const double rate = total_rate / n_rep;
+  if (rate <= 0.0 || rate > lam_max * 1000.0) return;
+
+  Stats fin = ev.eval(rate / 1000.0);
+  double rho = fin.n_serv / (double)p.max_batch;
+  rho = fmin(fmax(rho, 0.0), 1.0);
+
+  if (tid == 0) {
+    res[R_FEASIBLE] = 1.0;
+    res[R_REPLICAS] = n_rep;
+    res[R_RATE_STAR] = rate_star;
+    res[R_ITL] = eval_itl_of(p, fin);
+    res[R_TTFT] = eval_ttft_of(p, fin);
+    res[R_RHO] = rho;
+  }
+}
+
+extern "C" __global__ void __launch_bounds__(384) wva_solve_kernel_384(
+    const double *__restrict__ prob, double *__restrict__ out, int n_problems,
+    int max_k) {
+  solve_body_spec<2>(prob, out, n_problems, max_k);
+}
+
+extern "C" __global__ void __launch_bounds__(896) wva_solve_kernel_896(
+    const double *__restrict__ prob, double *__restrict__ out, int n_problems,
+    int max_k) {
+  solve_body_spec<3>(prob, out, n_problems, max_k);
+}
+
 extern "C" __global__ void __launch_bounds__(256) wva_solve_kernel_256(
     const double *__restrict__ prob, double *__restrict__ out, int n_problems,
     int max_k) {
@@ -419,28 +666,49 @@ extern "C" void wva_launch_solve(const double *prob, double *out, int n_problems
   // overrides (64 = the sequential single-wave kernel).
   const char *env = std::getenv("WVA_GPU_THREADS");
   int threads;
-  if (env != nullptr && std::strcmp(env, "256") == 0) {
+  // 64 KiB LDS per workgroup: cum[max_k] + totals[threads] + 64 slack
+  const int lds_doubles = 64 * 1024 / 8;
+  auto fits = [&](int thr) { return max_k + thr + 64 <= lds_doubles; };
+  if (env != nullptr && std::strcmp(env, "896") == 0 && fits(896)) {
+    threads = 896;
+  } else if (env != nullptr && std::strcmp(env, "384") == 0 && fits(384)) {
+    threads = 384;
+  } else if (env != nullptr && std::strcmp(env, "256") == 0) {
     threads = 256;
   } else if (env != nullptr && std::strcmp(env, "128") == 0) {
     threads = 128;
   } else if (env != nullptr && std::strcmp(env, "64") == 0) {
     threads = 64;
-  } else if (n_problems >= 512) {
-    // chip-filling launches: the dual kernel won or tied every measured
-    // (B, K) point at B=4096 (incl. K=7700, where it matches 4-wave)
+  } else if (n_problems >= 512 && max_k + 128 + 64 <= lds_doubles / 2) {
+    // Chip-filling launches with >= 2 concurrent 128-thread workgroups
+    // per CU are THROUGHPUT-bound: the dual kernel's 2 evaluations per
+    // round are the least redundant work (B=4096 rows of
+    // profiles/r02_kernel_spec.json: 0.44/0.77 ms vs spec's 0.77/1.13
+    // at K=704/2816).
     threads = 128;
   } else {
-    // underfilled launches are per-problem latency-bound: overlapping
-    // the two bisections wins while evaluations are short, but from
-    // K~2816 the 4-wave sweep parallelism matters more (B=192 data)
-    threads = max_k < 2048 ? 128 : 256;
+    // Latency-bound: small launches can't fill 256 CUs, and chains past
+    // half the LDS budget allow only one workgroup per CU even at
+    // B=4096 — either way the serial bisection IS the runtime, and the
+    // speculative-tree multisection kernel (depth 2, 2 x 3 waves,
+    // 2 levels per round) wins every measured point: 0.137/0.198/0.364
+    // ms at B=192 K=704/2816/7700 (vs 0.175/0.277/0.366 for the best
+    // r01 geometry) and 1.54 vs 2.02 ms at B=4096 K=7700
+    // (profiles/r02_kernel_spec.json).
+    threads = fits(384) ? 384 : 256;
   }
-  const size_t smem = (size_t)(max_k + threads + 32) * sizeof(double);
+  const size_t smem = (size_t)(max_k + threads + 64) * sizeof(double);
   if (threads == 64) {
     hipLaunchKernelGGL(wva::wva_solve_kernel_64, dim3(n_problems), dim3(64), smem,
                        (hipStream_t)stream, prob, out, n_problems, max_k);
   } else if (threads == 128) {
     hipLaunchKernelGGL(wva::wva_solve_kernel_128, dim3(n_problems), dim3(128), smem,
+                       (hipStream_t)stream, prob, out, n_problems, max_k);
+  } else if (threads == 384) {
+    hipLaunchKernelGGL(wva::wva_solve_kernel_384, dim3(n_problems), dim3(384), smem,
+                       (hipStream_t)stream, prob, out, n_problems, max_k);
+  } else if (threads == 896) {
+    hipLaunchKernelGGL(wva::wva_solve_kernel_896, dim3(n_problems), dim3(896), smem,
                        (hipStream_t)stream, prob, out, n_problems, max_k);
   } else {
     hipLaunchKernelGGL(wva::wva_solve_kernel_256, dim3(n_problems), dim3(256), smem,
