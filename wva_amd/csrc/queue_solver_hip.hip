@@ -50,6 +50,37 @@ namespace wva {
 // Reduction scratch layout (doubles, after cum[max_k] in dynamic LDS):
 //   red[0..WAVES*5-1]  per-wave partials (S, Ni, Snum, Ninum, eK)
 
+// Wave-parallel lower bound: smallest i in [lo, hi] with pred(i) true,
+// for a monotone (false...false true...true) predicate with pred(hi)
+// guaranteed true.  64 probes per round shrink the range 64x, replacing
+// ~11 DEPENDENT binary-search iterations (each an LDS-latency chain)
+// with ~2 ballot rounds of PARALLEL probes.  Exact: the boundary of a
+// monotone predicate is unique, so any probe strategy lands on the same
+// index the sequential bisection finds.
+template <typename Pred>
+__device__ inline int wave_lower_bound(int lo, int hi, int lane, Pred pred) {
+  while (hi - lo >= 64) {
+    const int step = (hi - lo + 63) >> 6;
+    const int i = lo + lane * step;  // lane 0 probes lo itself
+    const bool t = (i >= hi) ? true : pred(i);
+    const unsigned long long mask = __ballot(t);
+    if (mask == 0ull) {
+      // every probe (all below hi) is false: boundary past the last one
+      lo = lo + 63 * step + 1;
+      continue;
+    }
+    const int first = __ffsll((long long)mask) - 1;
+    if (first == 0) return lo;  // pred(lo) already true
+    lo = lo + (first - 1) * step + 1;
+    const int cap = lo - 1 + step;  // == previous lo + first*step
+    if (cap < hi) hi = cap;
+  }
+  const int i = lo + lane;
+  const bool t = (i >= hi) ? true : pred(i);
+  const unsigned long long mask = __ballot(t);
+  return lo + __ffsll((long long)mask) - 1;
+}
+
 template <int THREADS>
 struct WgEval {
   static constexpr int WAVES = THREADS / 64;
@@ -57,6 +88,33 @@ struct WgEval {
   const double *cum;  // LDS, K entries
   double *red;        // LDS scratch
   int K;
+
+  // Wave-parallel equivalents of log_mode_state/state_window
+  // (queue_core.h): same boundary indices, found with ballots instead of
+  // per-lane serial bisections.
+  __device__ int mode_state_wave(double lam, int lane) const {
+    double tN = prefill_time(p, (double)p.max_batch) +
+                p.num_decode * decode_time(p, (double)p.max_batch);
+    if ((double)p.max_batch <= lam * tN) return K;
+    double t1 = prefill_time(p, 1.0) + p.num_decode * decode_time(p, 1.0);
+    if (lam * t1 <= 1.0) return 0;
+    const int b = wave_lower_bound(
+        1, p.max_batch, lane, [&](int i) { return !rate_below(p, i, lam); });
+    return b - 1;
+  }
+
+  __device__ void window_wave(double loglam, int n_star, double m, int lane,
+                              int *lo_out, int *hi_out) const {
+    const double thresh = m - kLogCutoff;
+    *lo_out = wave_lower_bound(
+        0, n_star, lane, [&](int i) { return log_p(cum, loglam, i) >= thresh; });
+    // right edge: first n in (n_star, K+1] below threshold, minus one
+    // (probes past K count as below, so pred(K+1) is true by definition)
+    const int first_below = wave_lower_bound(
+        n_star, K + 1, lane,
+        [&](int i) { return i > K || log_p(cum, loglam, i) < thresh; });
+    *hi_out = first_below - 1;
+  }
 
   __device__ Stats eval(double lam) const {
     const int tid = threadIdx.x;
@@ -67,14 +125,22 @@ struct WgEval {
     const double loglam = log(lam);
 
     // max of logp via the concavity closed form, and the significant
-    // state window via two bisections (queue_core.h): every thread
-    // computes the identical O(log K) searches — no sweep, no barrier,
-    // and the exp sweep shrinks to the ~window where p(n) > 1e-16
-    const int n_star = log_mode_state(p, K, lam);
-    double m = log_p(cum, loglam, n_star);
-    if (m < 0.0) m = 0.0;
-    int n_lo, n_hi;
-    state_window(cum, loglam, K, n_star, m, &n_lo, &n_hi);
+    // state window via two boundary searches — wave-parallel (ballot)
+    // in the single-wave geometry, scalar per-thread otherwise; the
+    // exp sweep then shrinks to the ~window where p(n) > 1e-16
+    int n_star, n_lo, n_hi;
+    double m;
+    if constexpr (WAVES == 1) {
+      n_star = mode_state_wave(lam, lane);
+      m = log_p(cum, loglam, n_star);
+      if (m < 0.0) m = 0.0;
+      window_wave(loglam, n_star, m, lane, &n_lo, &n_hi);
+    } else {
+      n_star = log_mode_state(p, K, lam);
+      m = log_p(cum, loglam, n_star);
+      if (m < 0.0) m = 0.0;
+      state_window(cum, loglam, K, n_star, m, &n_lo, &n_hi);
+    }
     if constexpr (WAVES > 1) {
       __syncthreads();  // red may still be read from a previous eval
     }
