@@ -154,15 +154,48 @@ struct WgEval {
     const int num = p.max_batch;
     const int idx = (WAVES == 1) ? lane : tid;
     double S = 0.0, Ni = 0.0, Snum = 0.0, Ninum = 0.0, eK = 0.0;
-    for (int n = n_lo + idx; n <= n_hi; n += THREADS) {
-      double e = exp(log_p(cum, loglam, n) - m);
-      S += e;
-      Ni += (double)n * e;
-      if (n <= num) {
-        Snum += e;
-        Ninum += (double)n * e;
+    // 2x-unrolled sweep: two independent exp chains in flight per lane
+    // hide the fp64 transcendental latency the PMC profile flagged
+    // (~34% issue-stall); accumulators are kept separate and combined
+    // after the loop so the FLOATING-POINT SUM ORDER PER LANE CHANGES —
+    // covered by the cross-geometry ulp-tolerance in the parity suite.
+    {
+      double S1 = 0.0, Ni1 = 0.0, Snum1 = 0.0, Ninum1 = 0.0, eK1 = 0.0;
+      int n = n_lo + idx;
+      for (; n + THREADS <= n_hi; n += 2 * THREADS) {
+        const int n2 = n + THREADS;
+        const double e = exp(log_p(cum, loglam, n) - m);
+        const double e2 = exp(log_p(cum, loglam, n2) - m);
+        S += e;
+        Ni += (double)n * e;
+        S1 += e2;
+        Ni1 += (double)n2 * e2;
+        if (n <= num) {
+          Snum += e;
+          Ninum += (double)n * e;
+        }
+        if (n2 <= num) {
+          Snum1 += e2;
+          Ninum1 += (double)n2 * e2;
+        }
+        if (n == K) eK = e;
+        if (n2 == K) eK1 = e2;
       }
-      if (n == K) eK = e;
+      if (n <= n_hi) {
+        const double e = exp(log_p(cum, loglam, n) - m);
+        S += e;
+        Ni += (double)n * e;
+        if (n <= num) {
+          Snum += e;
+          Ninum += (double)n * e;
+        }
+        if (n == K) eK = e;
+      }
+      S += S1;
+      Ni += Ni1;
+      Snum += Snum1;
+      Ninum += Ninum1;
+      eK += eK1;
     }
     for (int off = 32; off > 0; off >>= 1) {
       S += __shfl_down(S, off, 64);
