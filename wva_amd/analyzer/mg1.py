@@ -24,15 +24,20 @@ This module provides the classical corrections:
   M/G/1 and a standard heuristic for the finite-capacity / state-
   dependent chains used by QueueAnalyzer.
 
-It is a capacity-planning refinement, deliberately NOT wired into the
-reconcile path: the controller keeps the reference's Markovian contract
-(an SLO sized under cs^2 = 1 is conservative for cs^2 < 1, which is the
-common case).  Use it offline to quantify that margin.
+Wiring (round 2): the correction is an OPT-IN analyzer mode.  Set
+``WVA_ANALYZER=mg1`` and ``WVA_SERVICE_SCV=<cs^2>`` (default 0.5 in mg1
+mode) and every sizing path — ``core.create_allocation`` and the batched
+fleet solver — constructs its QueueAnalyzer with that scv, scaling the
+predicted waiting times by ``(1 + cs^2)/2`` inside the Size bisections
+too.  The default remains the reference's Markovian contract (cs^2 = 1,
+identity); sizing under cs^2 = 1 when the workload is near-deterministic
+over-provisions — docs/design/mg1-analyzer.md carries measured margins.
 """
 
 from __future__ import annotations
 
 import math
+import os
 from dataclasses import dataclass
 
 __all__ = [
@@ -40,7 +45,30 @@ __all__ = [
     "service_scv_from_tokens",
     "MG1Corrector",
     "MG1Metrics",
+    "configured_scv",
 ]
+
+
+def configured_scv() -> float:
+    """The service-time cs^2 selected by environment, read per call so
+    tests and operators can flip modes without restarts.
+
+    WVA_ANALYZER: "" / "mm1k" (default, Markovian) or "mg1";
+    WVA_SERVICE_SCV: cs^2 override (mg1 mode defaults to 0.5 — halfway
+    between deterministic and exponential — when unset).
+    """
+    mode = os.environ.get("WVA_ANALYZER", "").strip().lower()
+    if mode in ("", "mm1k", "mm1"):
+        return 1.0
+    if mode != "mg1":
+        raise ValueError(f"unknown WVA_ANALYZER mode {mode!r} (use 'mm1k' or 'mg1')")
+    raw = os.environ.get("WVA_SERVICE_SCV", "").strip()
+    if not raw:
+        return 0.5
+    scv = float(raw)
+    if scv < 0:
+        raise ValueError(f"WVA_SERVICE_SCV must be >= 0, got {scv}")
+    return scv
 
 
 def pollaczek_khinchine_wait(arrival_rate: float, service_time: float, scv: float) -> float:

@@ -166,15 +166,30 @@ def build_service_rates(config: Configuration, request_size: RequestSize) -> np.
 
 class QueueAnalyzer:
     """Analyzer of one inference-server queue for a fixed (model, accelerator,
-    request-shape) point."""
+    request-shape) point.
 
-    def __init__(self, config: Configuration, request_size: RequestSize) -> None:
+    ``scv`` (squared coefficient of variation of service time) selects the
+    queueing model family: 1.0 (default) is the reference's Markovian
+    M/M/1/K contract; any other value applies the Allen-Cunneen wait
+    scaling ``(1 + cs^2)/2`` (exact for M/G/1) to every waiting-time
+    prediction — inside the Size bisections too, so lambda* and replica
+    counts respond.  Opt in fleet-wide with ``WVA_ANALYZER=mg1`` +
+    ``WVA_SERVICE_SCV`` (see analyzer/mg1.py for the theory and
+    docs/design/mg1-analyzer.md for measured margins)."""
+
+    def __init__(
+        self, config: Configuration, request_size: RequestSize, scv: float = 1.0
+    ) -> None:
         config.check()
         request_size.check()
+        if scv < 0:
+            raise AnalyzerError(f"invalid service-time scv {scv}")
         self.max_batch_size = config.max_batch_size
         self.max_queue_size = config.max_queue_size
         self.service_parms = config.service_parms
         self.request_size = request_size
+        self.scv = scv
+        self._wait_scale = (1.0 + scv) / 2.0
 
         serv_rate = build_service_rates(config, request_size)
         self.serv_rate = serv_rate
@@ -196,7 +211,7 @@ class QueueAnalyzer:
         eff_conc = effective_concurrency(
             m.avg_serv_time, self.service_parms, self.request_size, self.max_batch_size
         )
-        return m.avg_wait_time + self.service_parms.prefill.prefill_time(
+        return m.avg_wait_time * self._wait_scale + self.service_parms.prefill.prefill_time(
             self.request_size.avg_input_tokens, eff_conc
         )
 
@@ -226,10 +241,18 @@ class QueueAnalyzer:
         )
         token_time = self.service_parms.decode.decode_time(eff_conc)
         rho = min(max(avg_num_in_serv / float(self.max_batch_size), 0.0), 1.0)
+        corrected_wait = m.avg_wait_time * self._wait_scale
+        # scv == 1 keeps the reference's exact Markovian resp time
+        # (including its wait>=0 clamp edge case)
+        resp = (
+            m.avg_resp_time
+            if self._wait_scale == 1.0
+            else m.avg_serv_time + corrected_wait
+        )
         return AnalysisMetrics(
             throughput=m.throughput * 1000.0,
-            avg_resp_time=m.avg_resp_time,
-            avg_wait_time=m.avg_wait_time,
+            avg_resp_time=resp,
+            avg_wait_time=corrected_wait,
             avg_num_in_serv=avg_num_in_serv,
             avg_prefill_time=prefill_time,
             avg_token_time=token_time,

@@ -177,7 +177,9 @@ def create_allocation(system: "System", server_name: str, acc_name: str) -> Opti
     )
     request_size = RequestSize(avg_input_tokens=int(load.avg_in_tokens), avg_output_tokens=K)
     try:
-        qa = QueueAnalyzer(config, request_size)
+        from ..analyzer.mg1 import configured_scv
+
+        qa = QueueAnalyzer(config, request_size, scv=configured_scv())
     except AnalyzerError:
         return None
 

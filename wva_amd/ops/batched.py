@@ -56,6 +56,7 @@ GPU_MAX_BATCH_LIMIT = 700
 _graph_cache: dict = {}
 _graph_lock = None
 _graph_disabled = False
+_mg1_warned = False
 
 
 def _graph_solve(native, problems_t, max_k):
@@ -108,7 +109,7 @@ def _graph_solve(native, problems_t, max_k):
             return None
 
 
-def _solve_problems_python(problems: np.ndarray) -> np.ndarray:
+def _solve_problems_python(problems: np.ndarray, scv: float = 1.0) -> np.ndarray:
     """Reference-semantics scalar fallback via the Python analyzer."""
     out = np.zeros((problems.shape[0], RESULT_FIELDS), dtype=np.float64)
     for i, row in enumerate(problems):
@@ -128,6 +129,7 @@ def _solve_problems_python(problems: np.ndarray) -> np.ndarray:
                     avg_input_tokens=int(row[P_IN_TOKENS]),
                     avg_output_tokens=int(row[P_OUT_TOKENS]),
                 ),
+                scv=scv,
             )
             _, metrics, _ = qa.size(
                 TargetPerf(
@@ -165,6 +167,25 @@ def solve_problems(problems: np.ndarray, device: Optional[str] = None) -> np.nda
         raise ValueError(f"problems must be [B, {PROBLEM_FIELDS}]")
     if problems.shape[0] == 0:
         return np.zeros((0, RESULT_FIELDS), dtype=np.float64)
+
+    from ..analyzer.mg1 import configured_scv
+
+    scv = configured_scv()
+    if scv != 1.0:
+        # WVA_ANALYZER=mg1: the Allen-Cunneen wait scaling lives in the
+        # Python analyzer; the native kernels implement the Markovian
+        # contract only, so mg1 mode routes through the scalar path
+        # (opt-in capacity-planning mode, not the hot default)
+        global _mg1_warned
+        if not _mg1_warned:
+            _mg1_warned = True
+            import logging
+
+            logging.getLogger("wva_amd.ops").info(
+                "WVA_ANALYZER=mg1 (cs^2=%.3g): sizing via the Python "
+                "M/G/1-corrected analyzer instead of the native kernel", scv,
+            )
+        return _solve_problems_python(problems, scv=scv)
 
     want_gpu = device is not None and str(device).startswith("cuda")
     if want_gpu:
