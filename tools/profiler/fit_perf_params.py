@@ -207,6 +207,59 @@ def _linfit(x: np.ndarray, y: np.ndarray) -> Tuple[float, float, float]:
     return float(coef[0]), float(coef[1]), r2
 
 
+def synthetic_caches(
+    model: TinyDecoder, B: int, kv_len: int, device: str, dtype, paged: bool,
+    page_size: int = 16,
+):
+    """Pre-filled KV caches of length ``kv_len`` (random-init — the
+    decode step's memory traffic, not the cache contents, is what's
+    measured).  With ``paged`` the cache is stored page-shuffled and
+    gathered through a page table each step, so the reads follow the
+    block-indirection pattern of paged attention rather than one
+    contiguous stream."""
+    heads = model.layers[0].heads
+    hd = model.layers[0].head_dim
+    caches = []
+    n_pages = (kv_len + page_size - 1) // page_size
+    perm = torch.randperm(n_pages * page_size, device=device)[:kv_len] if paged else None
+    for _ in model.layers:
+        k = torch.randn(B, heads, kv_len, hd, device=device, dtype=dtype)
+        v = torch.randn(B, heads, kv_len, hd, device=device, dtype=dtype)
+        caches.append((k, v))
+    return caches, perm
+
+
+def decode_step_kv(model: TinyDecoder, tokens, caches, perm):
+    """One decode step over the synthetic caches; with a page table the
+    K/V reads go through index_select (the gather a paged-KV kernel
+    performs), without it they are the contiguous concat path."""
+    import torch.nn.functional as F
+
+    x = model.embed(tokens)
+    for layer, (k_cache, v_cache) in zip(model.layers, caches):
+        B, T, H = x.shape
+        residual = x
+        h = layer.norm1(x)
+        qkv = layer.qkv(h).view(B, T, 3, layer.heads, layer.head_dim)
+        q, k, v = qkv.unbind(2)
+        q, k, v = (t.transpose(1, 2) for t in (q, k, v))
+        if perm is not None:
+            k_read = k_cache.index_select(2, perm)
+            v_read = v_cache.index_select(2, perm)
+        else:
+            k_read, v_read = k_cache, v_cache
+        k = torch.cat([k_read, k], dim=2)
+        v = torch.cat([v_read, v], dim=2)
+        attn = F.scaled_dot_product_attention(q, k, v)
+        h = layer.o(attn.transpose(1, 2).reshape(B, T, H))
+        x = residual + h
+        if hasattr(layer, "moe"):
+            x = x + layer.moe(layer.norm2(x))
+        else:
+            x = x + layer.down(F.silu(layer.up(layer.norm2(x))))
+    return model.head(x)
+
+
 def measure(
     model: TinyDecoder,
     device: str,
@@ -214,11 +267,14 @@ def measure(
     seq_len: int,
     decode_iters: int,
     warmup: int,
+    kv_len: int = 0,
+    paged: bool = False,
 ) -> FitResult:
     def sync():
         if device.startswith("cuda"):
             torch.cuda.synchronize()
 
+    dtype = next(model.parameters()).dtype
     decode_points = []
     prefill_points = []
     with torch.no_grad():
@@ -235,17 +291,33 @@ def measure(
             prefill_ms = (time.perf_counter() - t0) / max(decode_iters // 4, 1) * 1000.0
             prefill_points.append((B * seq_len, prefill_ms))
 
-            # decode timing with KV cache
+            # decode timing with KV cache: either the caches the prefill
+            # produced (short-context; weight-bound) or synthetic
+            # long-context caches (kv_len > 0; KV-bandwidth-bound, the
+            # regime real paged-attention decode lives in)
+            perm = None
+            if kv_len > 0:
+                del caches
+                caches, perm = synthetic_caches(model, B, kv_len, device, dtype, paged)
             step_tokens = torch.randint(0, 31999, (B, 1), device=device)
+
+            def step():
+                if kv_len > 0:
+                    decode_step_kv(model, step_tokens, caches, perm)
+                else:
+                    model.decode_step(step_tokens, caches)
+
             for _ in range(warmup):
-                model.decode_step(step_tokens, caches)
+                step()
             sync()
             t0 = time.perf_counter()
             for _ in range(decode_iters):
-                model.decode_step(step_tokens, caches)
+                step()
             sync()
             decode_ms = (time.perf_counter() - t0) / decode_iters * 1000.0
             decode_points.append((B, decode_ms))
+            if kv_len > 0:
+                del caches  # free before the next batch size
 
     bx = np.array([p[0] for p in decode_points], dtype=np.float64)
     by = np.array([p[1] for p in decode_points], dtype=np.float64)
@@ -269,6 +341,8 @@ def fit(
     dtype=torch.bfloat16,
     fp8: bool = False,
     moe_experts: int = 0,
+    kv_len: int = 0,
+    paged: bool = False,
 ) -> FitResult:
     device = device or ("cuda" if torch.cuda.is_available() else "cpu")
     if device == "cpu":
@@ -282,11 +356,19 @@ def fit(
             if isinstance(module, FP8Linear):
                 module.weight_fp8 = module.weight_fp8.to(torch.float8_e4m3fn)
                 module.weight_scale = module.weight_scale.float()
-    return fit_with_model(model, device, list(batches), seq_len, decode_iters, warmup)
+    return fit_with_model(
+        model, device, list(batches), seq_len, decode_iters, warmup,
+        kv_len=kv_len, paged=paged,
+    )
 
 
-def fit_with_model(model, device, batches, seq_len, decode_iters, warmup) -> FitResult:
-    return measure(model, device, batches, seq_len, decode_iters, warmup)
+def fit_with_model(
+    model, device, batches, seq_len, decode_iters, warmup, kv_len=0, paged=False
+) -> FitResult:
+    return measure(
+        model, device, batches, seq_len, decode_iters, warmup,
+        kv_len=kv_len, paged=paged,
+    )
 
 
 def main() -> None:
@@ -302,6 +384,13 @@ def main() -> None:
     ap.add_argument("--dtype", choices=["bf16", "fp8"], default="bf16")
     ap.add_argument("--moe-experts", type=int, default=0,
                     help="top-2 MoE with this many experts per layer (0 = dense)")
+    ap.add_argument("--kv-len", type=int, default=0,
+                    help="synthetic KV-cache length for decode timing (0 = "
+                         "use the prefill's short cache); large values make "
+                         "the decode step KV-bandwidth bound")
+    ap.add_argument("--paged", action="store_true",
+                    help="read the KV cache through a shuffled page table "
+                         "(paged-attention gather pattern)")
     ap.add_argument("--out", default="")
     args = ap.parse_args()
 
@@ -315,11 +404,15 @@ def main() -> None:
         warmup=args.warmup,
         fp8=args.dtype == "fp8",
         moe_experts=args.moe_experts,
+        kv_len=args.kv_len,
+        paged=args.paged,
     )
     payload = {
         "acc": args.acc,
         "model": f"tiny-decoder-L{args.layers}-H{args.hidden}-{args.dtype}"
-        + (f"-moe{args.moe_experts}x" if args.moe_experts else ""),
+        + (f"-moe{args.moe_experts}x" if args.moe_experts else "")
+        + (f"-kv{args.kv_len}" if args.kv_len else "")
+        + ("-paged" if args.paged else ""),
         "perfParms": {
             "decodeParms": {"alpha": f"{result.alpha:.4f}", "beta": f"{result.beta:.6f}"},
             "prefillParms": {"gamma": f"{result.gamma:.4f}", "delta": f"{result.delta:.8f}"},
