@@ -190,8 +190,11 @@ class TestWatch:
         try:
             runtime = ManagerRuntime(client, prom_api=MockPromAPI())
             assert runtime._watch_threads  # HTTP tier gets watcher threads
-            assert not runtime._wake.is_set()
-            time.sleep(0.3)
+            # the session's initial list may fire a startup wake for the
+            # pre-existing watched ConfigMap (informer initial-sync Add
+            # semantics, same as controller-runtime); absorb it
+            time.sleep(0.5)
+            runtime._wake.clear()
             make_va(store, name="wake-va")
             for _ in range(60):
                 if runtime._wake.is_set():
