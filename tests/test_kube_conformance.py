@@ -167,6 +167,59 @@ class TestSchemaValidation:
             v.validate(va, subresource="status")
 
 
+class TestAdmissionValidationMatrix:
+    """The reference's envtest admission scenarios
+    (variantautoscaling_controller_test.go:411-533), run against the
+    stub's server-side schema validation."""
+
+    def _post(self, client, body):
+        return client._client.post(
+            "/apis/llmd.ai/v1alpha1/namespaces/default/variantautoscalings",
+            json=body,
+        )
+
+    def _body(self, name):
+        return valid_va(name).model_dump(by_alias=True, exclude_none=True, mode="json")
+
+    def test_validates_accelerator_profiles(self, api):
+        # :411 negative accCount / maxBatchSize rejected at the API level
+        client, _ = api
+        body = self._body("invalid-profile")
+        acc = body["spec"]["modelProfile"]["accelerators"][0]
+        acc["accCount"] = -1
+        acc["maxBatchSize"] = -1
+        resp = self._post(client, body)
+        assert resp.status_code == 422
+        assert "accCount" in resp.json()["message"]
+
+    def test_handles_empty_model_id(self, api):
+        # :444 empty ModelID -> error names spec.modelID
+        client, _ = api
+        body = self._body("invalid-model-id")
+        body["spec"]["modelID"] = ""
+        resp = self._post(client, body)
+        assert resp.status_code == 422
+        assert "modelID" in resp.json()["message"]
+
+    def test_handles_empty_accelerator_list(self, api):
+        # :477 no accelerators -> error names the accelerators field
+        client, _ = api
+        body = self._body("empty-accelerators")
+        body["spec"]["modelProfile"]["accelerators"] = []
+        resp = self._post(client, body)
+        assert resp.status_code == 422
+        assert "accelerators" in resp.json()["message"]
+
+    def test_handles_empty_slo_class_ref(self, api):
+        # :502 empty SLOClassRef -> error names sloClassRef
+        client, _ = api
+        body = self._body("empty-slo-class-ref")
+        body["spec"]["sloClassRef"] = {}
+        resp = self._post(client, body)
+        assert resp.status_code == 422
+        assert "sloClassRef" in resp.json()["message"]
+
+
 class TestChunkedList:
     def test_pagination_with_continue(self, api):
         client, store = api
