@@ -98,3 +98,50 @@ class TestReplicaSource:
             )
             == 1.0
         )
+
+
+class TestRatioScenarioTable:
+    """actuator_test.go:585 'ratio calculation scenarios' + the metrics
+    integration context, as a table over the emitter directly."""
+
+    @pytest.mark.parametrize(
+        "current,desired,want_ratio",
+        [
+            (2, 5, 2.5),     # scale-up
+            (0, 3, 3.0),     # 0 -> N encoded as ratio = N (metrics.go:118-124)
+            (4, 4, 1.0),     # no change
+            (6, 2, 2 / 6),   # scale-down
+            (0, 0, 0.0),     # idle at zero (scale-to-zero steady state)
+        ],
+    )
+    def test_ratio(self, registry, current, desired, want_ratio):
+        va = make_va(desired=desired, current=current)
+        ctrl_metrics.MetricsEmitter().emit_replica_metrics(va, current, desired, "MI355X")
+        assert gauge(registry, "inferno_desired_ratio", va) == pytest.approx(want_ratio)
+        assert gauge(registry, "inferno_desired_replicas", va) == desired
+        assert gauge(registry, "inferno_current_replicas", va) == current
+
+    def test_emit_metrics_zero_desired_still_emits(self, registry):
+        # actuator.go:50-84: the guard is `>= 0` — desired == 0 still
+        # emits (scale-to-zero depends on it)
+        client = InMemoryKubeClient()
+        act = Actuator(client)
+        va_zero = make_va(name="zero", desired=0, current=1)
+        act.emit_metrics(va_zero)  # deployment missing -> fallback path
+        assert gauge(registry, "inferno_desired_replicas", va_zero) == 0
+
+    def test_negative_desired_rejected_at_the_type_level(self):
+        # the reference skips emission for negative desired at runtime;
+        # here the typed model (minimum: 0, CRD parity) makes a negative
+        # count unrepresentable — stronger than the runtime guard
+        with pytest.raises(Exception):
+            v1alpha1.OptimizedAlloc(accelerator="MI355X", numReplicas=-1)
+
+    def test_missing_deployment_uses_va_status_fallback(self, registry):
+        # actuator_test.go:274 fallback replicas when retrieval fails
+        client = InMemoryKubeClient()
+        act = Actuator(client)
+        va = make_va(name="orphan", desired=5, current=3)
+        act.emit_metrics(va)  # no Deployment anywhere
+        assert gauge(registry, "inferno_current_replicas", va) == 3
+        assert gauge(registry, "inferno_desired_replicas", va) == 5
