@@ -76,3 +76,91 @@ class TestInventoryStub:
     def test_stub_and_vendor_order(self):
         assert collector.collect_inventory_k8s(None) == {}
         assert collector.VENDORS[0] == "amd.com"  # AMD first-class
+
+
+class TestValidateMetricsAvailability:
+    """collector_test.go:410-530 validation scenarios, as a table."""
+
+    MODEL, NS = "m/llama", "prod"
+
+    def _q(self, with_ns=True):
+        from wva_amd.controller import constants
+
+        if with_ns:
+            return (
+                f'{constants.VLLM_REQUEST_SUCCESS_TOTAL}'
+                f'{{{constants.LABEL_MODEL_NAME}="{self.MODEL}",'
+                f'{constants.LABEL_NAMESPACE}="{self.NS}"}}'
+            )
+        return (
+            f'{constants.VLLM_REQUEST_SUCCESS_TOTAL}'
+            f'{{{constants.LABEL_MODEL_NAME}="{self.MODEL}"}}'
+        )
+
+    def test_available_with_namespace_label(self):
+        prom = MockPromAPI()
+        prom.set_result(self._q(), 5.0)
+        r = collector.validate_metrics_availability(prom, self.MODEL, self.NS)
+        assert r.available
+
+    def test_fallback_without_namespace_label(self):
+        # emulator scrape: first query empty, fallback (no namespace)
+        # has the series -> available
+        prom = MockPromAPI()
+        prom.query_results[self._q()] = []
+        prom.set_result(self._q(with_ns=False), 5.0)
+        r = collector.validate_metrics_availability(prom, self.MODEL, self.NS)
+        assert r.available
+
+    def test_unavailable_on_prometheus_error(self):
+        prom = MockPromAPI()
+        prom.set_error(self._q(), RuntimeError("boom"))
+        r = collector.validate_metrics_availability(prom, self.MODEL, self.NS)
+        assert not r.available
+        assert r.reason == v1alpha1.REASON_PROMETHEUS_ERROR
+
+    def test_unavailable_when_no_metrics_found(self):
+        prom = MockPromAPI()
+        prom.query_results[self._q()] = []
+        prom.query_results[self._q(with_ns=False)] = []
+        r = collector.validate_metrics_availability(prom, self.MODEL, self.NS)
+        assert not r.available
+        assert r.reason == v1alpha1.REASON_METRICS_MISSING
+        # the message carries the operator checklist
+        assert "ServiceMonitor" in r.message
+
+    def test_unavailable_when_metrics_stale(self):
+        prom = MockPromAPI()
+        prom.set_result(self._q(), 5.0, age_seconds=301.0)  # > 5 min
+        r = collector.validate_metrics_availability(prom, self.MODEL, self.NS)
+        assert not r.available
+        assert r.reason == v1alpha1.REASON_METRICS_STALE
+
+    def test_fallback_query_error_is_prometheus_error(self):
+        prom = MockPromAPI()
+        prom.query_results[self._q()] = []
+        prom.set_error(self._q(with_ns=False), RuntimeError("fallback boom"))
+        r = collector.validate_metrics_availability(prom, self.MODEL, self.NS)
+        assert not r.available
+        assert r.reason == v1alpha1.REASON_PROMETHEUS_ERROR
+
+    def test_fresh_metrics_within_window_accepted(self):
+        prom = MockPromAPI()
+        prom.set_result(self._q(), 5.0, age_seconds=299.0)  # just inside
+        r = collector.validate_metrics_availability(prom, self.MODEL, self.NS)
+        assert r.available
+
+
+class TestFixValueTable:
+    """collector_test.go:372-408 FixValue table."""
+
+    def test_table(self):
+        import math
+
+        fix = collector._fix_value
+        assert fix(float("nan")) == 0.0
+        assert fix(float("inf")) == 0.0
+        assert fix(float("-inf")) == 0.0
+        assert fix(42.5) == 42.5
+        assert fix(0.0) == 0.0
+        assert fix(-3.25) == -3.25  # negatives pass through unchanged
