@@ -465,3 +465,37 @@ class TestSolveCLI:
         spec = make_spec()
         result = solve_spec(load_spec(spec.to_json()))
         assert "s1:default" in result["allocations"]
+
+
+class TestManagerEdgeCases:
+    """manager_test.go:334 TestManager_EdgeCases — invalid wiring must
+    surface as an error, never silently no-op."""
+
+    def test_nil_system_errors(self):
+        from wva_amd.config import OptimizerSpec
+
+        manager = Manager(None, Optimizer(OptimizerSpec(unlimited=True)))
+        with pytest.raises(Exception):
+            manager.optimize()
+
+    def test_nil_optimizer_errors(self):
+        system, _ = make_system()
+        manager = Manager(system, None)
+        with pytest.raises(Exception):
+            manager.optimize()
+
+    def test_optimize_produces_solution_and_type_aggregation(self):
+        # manager_test.go:61 TestManager_Optimize happy path: optimize,
+        # then the per-type aggregation and solution export line up
+        system, opt = make_system(
+            servers=[server_spec("m1:ns", arrival_rate=120.0)]
+        )
+        system.calculate()
+        Manager(system, Optimizer(opt)).optimize()
+        solution = system.generate_solution()
+        assert "m1:ns" in solution.spec
+        entry = solution.spec["m1:ns"]
+        assert entry.num_replicas >= 1
+        by_type = system.allocation_by_type
+        total_units = sum(a.count for a in by_type.values())
+        assert total_units >= entry.num_replicas
