@@ -118,3 +118,66 @@ class TestCRDSchemaDoc:
         import gen_crd_docs
 
         assert gen_crd_docs.DOC_PATH.read_text() == gen_crd_docs.render()
+
+
+class TestDeepCopyIndependence:
+    """variantautoscaling_types_test.go:97 — mutated copies must never
+    alias the original's nested fields (pydantic model_copy(deep=True)
+    stands in for the generated DeepCopy)."""
+
+    def _valid_va(self):
+        return v1alpha1.VariantAutoscaling.model_validate({
+            "metadata": {"name": "a", "namespace": "ns"},
+            "spec": {
+                "modelID": "model-123",
+                "sloClassRef": {"name": "slo-config", "key": "k"},
+                "modelProfile": {"accelerators": [{
+                    "acc": "MI355X", "accCount": 1, "maxBatchSize": 8,
+                    "perfParms": {"decodeParms": {"alpha": "1", "beta": "2"},
+                                  "prefillParms": {"gamma": "3", "delta": "4"}},
+                }]},
+            },
+        })
+
+    def test_deep_copy_independence(self):
+        orig = self._valid_va()
+        cp = orig.model_copy(deep=True)
+        cp.spec.model_id = "model-456"
+        cp.spec.slo_class_ref.name = "slo-config-2"
+        cp.spec.model_profile.accelerators[0].acc = "MI300X"
+        cp.status.current_alloc.load.arrival_rate = "20"
+        assert orig.spec.model_id == "model-123"
+        assert orig.spec.slo_class_ref.name == "slo-config"
+        assert orig.spec.model_profile.accelerators[0].acc == "MI355X"
+        assert orig.status.current_alloc.load.arrival_rate != "20"
+
+    def test_json_round_trip_exact(self):
+        # :120 TestJSONRoundTrip incl. the RFC3339 lastRunTime instant
+        import datetime
+        import json
+
+        orig = self._valid_va()
+        orig.status.desired_optimized_alloc.accelerator = "MI355X"
+        orig.status.desired_optimized_alloc.last_run_time = datetime.datetime(
+            2026, 9, 14, 12, 0, 0, tzinfo=datetime.timezone.utc
+        )
+        raw = json.dumps(orig.model_dump(by_alias=True, exclude_none=True, mode="json"))
+        back = v1alpha1.VariantAutoscaling.model_validate(json.loads(raw))
+        assert (
+            back.status.desired_optimized_alloc.last_run_time
+            == orig.status.desired_optimized_alloc.last_run_time
+        )
+        assert back.model_dump(by_alias=True, exclude_none=True, mode="json") == \
+            orig.model_dump(by_alias=True, exclude_none=True, mode="json")
+
+    def test_last_run_time_serializes_rfc3339(self):
+        # :228 TestOptimizedAllocLastRunTimeJSON
+        import datetime
+
+        va = self._valid_va()
+        va.status.desired_optimized_alloc.last_run_time = datetime.datetime(
+            2026, 9, 14, 12, 0, 0, tzinfo=datetime.timezone.utc
+        )
+        d = va.model_dump(by_alias=True, exclude_none=True, mode="json")
+        raw = d["status"]["desiredOptimizedAlloc"]["lastRunTime"]
+        assert isinstance(raw, str) and raw.startswith("2026-09-14T12:00:00")
