@@ -337,11 +337,14 @@ class CreateWatchSession:
 
         On first prime this surfaces every existing object — the same
         initial-sync Add events a controller-runtime informer delivers.
+        A prime also compacts the seen-uid set to the live objects, so
+        long sessions on churny clusters stay bounded (deleted objects'
+        uids would otherwise accumulate forever).
         """
         items, rv = self.client.list_with_rv(self.cls, self.namespace)
         self._rv = rv
         fresh = [o for o in items if o.metadata.uid not in self._seen_uids]
-        self._seen_uids.update(o.metadata.uid for o in items)
+        self._seen_uids = {o.metadata.uid for o in items}
         return fresh
 
     def run(self, callback) -> None:
@@ -365,6 +368,8 @@ class CreateWatchSession:
                         self._seen_uids.add(obj.metadata.uid)
                         callback(obj)
                 self.backoff_s = self.BACKOFF_BASE_S
+                if len(self._seen_uids) > 65536:
+                    primed = False  # force a compacting re-list
             except GoneError:
                 # watch history compacted: re-list immediately, resume
                 # from the fresh collection rv
