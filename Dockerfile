@@ -8,7 +8,7 @@ FROM python:3.10-slim AS build
 
 RUN apt-get update && apt-get install -y --no-install-recommends g++ \
     && rm -rf /var/lib/apt/lists/* \
-    && pip install --no-cache-dir pybind11 setuptools numpy
+    && pip install --no-cache-dir pybind11==3.0.4 setuptools numpy==2.2.6
 
 WORKDIR /src
 COPY setup.py ./

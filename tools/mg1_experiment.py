@@ -21,7 +21,6 @@ import argparse
 import asyncio
 import json
 import sys
-import threading
 import time
 from pathlib import Path
 

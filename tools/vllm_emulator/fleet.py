@@ -16,7 +16,7 @@ from __future__ import annotations
 import asyncio
 import threading
 import time
-from typing import List, Optional
+from typing import List
 
 from .engine import EmulatorSettings
 from .server import create_app
