@@ -178,6 +178,47 @@ WVA_HD void state_window(const double *cum, double loglam, int K, int n_star,
   *hi_out = lo;
 }
 
+// Closed-form geometric tail of the state sums (the "queue region").
+//
+// For n >= N-1 the service rate is constant (mu(n) = serv_rate(N)), so
+// consecutive log-probabilities differ by the constant d = log(lam) -
+// log(mu_N) and the normalized probabilities e^{logp(n)-m} form a
+// geometric sequence.  Summing states [g_start, n_hi] (g_start > N, so
+// none of them contribute to the <=N partial sums) in closed form
+// replaces up to 10*N per-state exp() calls — the dominant fp64 cost of
+// an evaluation — with a handful of exp/expm1.
+//
+//   S_geo  = p_a * G0,            G0 = sum_{j=0..M} r^j = expm1((M+1)d)/expm1(d)
+//   Ni_geo = p_a * (a*G0 + G1),   G1 = sum_{j=0..M} j r^j
+//                                    = r (1 - (M+1) r^M + M r^{M+1}) / (1-r)^2
+//
+// Near r == 1 the G1 numerator cancels; |d|*(M+1) < 0.01 falls back to
+// the term-wise loop (a one-or-two-iterate sliver of the bisection).
+struct GeoTail {
+  double S, Ni, eK;
+  bool used;
+};
+
+WVA_HD GeoTail geo_tail(const double *cum, double loglam, double d, double m,
+                        int g_start, int n_hi, int K) {
+  GeoTail t{0.0, 0.0, 0.0, false};
+  const int M = n_hi - g_start;
+  if (M < 8 || fabs(d) * (double)(M + 1) < 0.01) return t;  // loop is fine/safer
+  const double x_a = log_p(cum, loglam, g_start) - m;
+  const double p_a = exp(x_a);
+  const double e1 = expm1(d);
+  const double G0 = expm1((double)(M + 1) * d) / e1;
+  const double r = e1 + 1.0;
+  const double rM = exp((double)M * d);
+  const double G1 =
+      r * (1.0 - (double)(M + 1) * rM + (double)M * rM * r) / (e1 * e1);
+  t.S = p_a * G0;
+  t.Ni = p_a * ((double)g_start * G0 + G1);
+  if (n_hi == K) t.eK = p_a * exp((double)(K - g_start) * d);
+  t.used = true;
+  return t;
+}
+
 // Scalar model evaluation at arrival rate lam (req/ms) given the inclusive
 // cumulative sum cum[n] = sum_{i<=n} log_mu(i), n in [0, K-1].
 WVA_HD Stats eval_model(const Parms &p, const double *cum, int K, double lam) {
@@ -190,7 +231,16 @@ WVA_HD Stats eval_model(const Parms &p, const double *cum, int K, double lam) {
   state_window(cum, loglam, K, n_star, m, &n_lo, &n_hi);
   int num = p.max_batch;  // serv_rate array length
   double S = 0.0, Ni = 0.0, Snum = 0.0, Ninum = 0.0, eK = 0.0;
-  for (int n = n_lo; n <= n_hi; ++n) {
+  // geometric tail over the constant-rate queue region (states > N)
+  const int g_start = (num + 1 > n_lo) ? num + 1 : n_lo;
+  GeoTail tail{0.0, 0.0, 0.0, false};
+  int loop_hi = n_hi;
+  if (g_start + 8 <= n_hi) {  // also keeps cum[g_start] in bounds
+    const double d = loglam - (cum[g_start] - cum[g_start - 1]);
+    tail = geo_tail(cum, loglam, d, m, g_start, n_hi, K);
+    if (tail.used) loop_hi = g_start - 1;
+  }
+  for (int n = n_lo; n <= loop_hi; ++n) {
     double e = exp(log_p(cum, loglam, n) - m);
     S += e;
     Ni += n * e;
@@ -200,6 +250,9 @@ WVA_HD Stats eval_model(const Parms &p, const double *cum, int K, double lam) {
     }
     if (n == K) eK = e;
   }
+  S += tail.S;
+  Ni += tail.Ni;
+  eK += tail.eK;
   Stats st;
   st.throughput = lam * (1.0 - eK / S);
   double n_sys = Ni / S;

@@ -153,6 +153,19 @@ struct WgEval {
     // own full window in lockstep, reductions stay wave-local).
     const int num = p.max_batch;
     const int idx = (WAVES == 1) ? lane : tid;
+    // Geometric tail over the constant-service-rate queue region
+    // (queue_core.h geo_tail): closed-form sum of the states past N,
+    // which are up to 10/11 of the window — the per-state exp calls
+    // they would cost are the kernel's dominant fp64 work.  Uniform
+    // across lanes; added to the reduced totals below.
+    const int g_start = (num + 1 > n_lo) ? num + 1 : n_lo;
+    GeoTail tail{0.0, 0.0, 0.0, false};
+    int loop_hi = n_hi;
+    if (g_start + 8 <= n_hi) {
+      const double d = loglam - (cum[g_start] - cum[g_start - 1]);
+      tail = geo_tail(cum, loglam, d, m, g_start, n_hi, K);
+      if (tail.used) loop_hi = g_start - 1;
+    }
     double S = 0.0, Ni = 0.0, Snum = 0.0, Ninum = 0.0, eK = 0.0;
     // 2x-unrolled sweep: two independent exp chains in flight per lane
     // hide the fp64 transcendental latency the PMC profile flagged
@@ -162,7 +175,7 @@ struct WgEval {
     {
       double S1 = 0.0, Ni1 = 0.0, Snum1 = 0.0, Ninum1 = 0.0, eK1 = 0.0;
       int n = n_lo + idx;
-      for (; n + THREADS <= n_hi; n += 2 * THREADS) {
+      for (; n + THREADS <= loop_hi; n += 2 * THREADS) {
         const int n2 = n + THREADS;
         const double e = exp(log_p(cum, loglam, n) - m);
         const double e2 = exp(log_p(cum, loglam, n2) - m);
@@ -181,7 +194,7 @@ struct WgEval {
         if (n == K) eK = e;
         if (n2 == K) eK1 = e2;
       }
-      if (n <= n_hi) {
+      if (n <= loop_hi) {
         const double e = exp(log_p(cum, loglam, n) - m);
         S += e;
         Ni += (double)n * e;
@@ -229,6 +242,10 @@ struct WgEval {
         eK += red[w * 5 + 4];
       }
     }
+    // geometric tail joins after the reduction (uniform on all lanes)
+    S += tail.S;
+    Ni += tail.Ni;
+    eK += tail.eK;
 
     Stats st;
     st.throughput = lam * (1.0 - eK / S);
