@@ -795,22 +795,20 @@ extern "C" void wva_launch_solve(const double *prob, double *out, int n_problems
     threads = 128;
   } else if (env != nullptr && std::strcmp(env, "64") == 0) {
     threads = 64;
-  } else if (n_problems >= 512 && max_k + 128 + 64 <= lds_doubles / 2) {
-    // Chip-filling launches with >= 2 concurrent 128-thread workgroups
-    // per CU are THROUGHPUT-bound: the dual kernel's 2 evaluations per
-    // round are the least redundant work (B=4096 rows of
-    // profiles/r02_kernel_spec.json: 0.44/0.77 ms vs spec's 0.77/1.13
-    // at K=704/2816).
+  } else if (n_problems >= 512) {
+    // Chip-filling launches are THROUGHPUT-bound: with the geometric
+    // queue-tail (queue_core.h geo_tail) collapsing the sweep cost, the
+    // dual kernel's 2 evaluations per round are the least redundant
+    // work at every measured chip-filling point — B=4096 rows of
+    // profiles/r02_kernel_geo.json: 0.33/0.44/1.04 ms at
+    // K=704/2816/7700 vs spec-384's 0.75/1.01/1.21.
     threads = 128;
   } else {
-    // Latency-bound: small launches can't fill 256 CUs, and chains past
-    // half the LDS budget allow only one workgroup per CU even at
-    // B=4096 — either way the serial bisection IS the runtime, and the
-    // speculative-tree multisection kernel (depth 2, 2 x 3 waves,
-    // 2 levels per round) wins every measured point: 0.137/0.198/0.364
-    // ms at B=192 K=704/2816/7700 (vs 0.175/0.277/0.366 for the best
-    // r01 geometry) and 1.54 vs 2.02 ms at B=4096 K=7700
-    // (profiles/r02_kernel_spec.json).
+    // Underfilled launches can't occupy 256 CUs, so the serial
+    // bisection IS the runtime and the speculative-tree multisection
+    // kernel (depth 2, 2 x 3 waves, 2 levels per round) wins every
+    // measured point: 0.106/0.116/0.141 ms at B=192 K=704/2816/7700
+    // (r01 best geometry: 0.175/0.277/0.360; profiles/r02_kernel_geo.json).
     threads = fits(384) ? 384 : 256;
   }
   const size_t smem = (size_t)(max_k + threads + 64) * sizeof(double);
