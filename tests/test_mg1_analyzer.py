@@ -177,3 +177,70 @@ class TestSizingRespondsToScv:
         r0, r1 = self._rate_star(0.0), self._rate_star(1.0)
         margin_pct = (r0 - r1) / r1 * 100.0
         assert margin_pct > 3.0
+
+
+class TestMG1ThroughReconciler:
+    def test_mg1_mode_lowers_wait_bound_replicas(self, monkeypatch):
+        """Fleet-level integration: with a wait-bound (tight-TTFT)
+        workload, the reconciler sizes fewer-or-equal replicas under
+        WVA_ANALYZER=mg1 (cs^2=0) than under the Markovian default,
+        end to end through the controller."""
+        import sys
+        from pathlib import Path
+
+        sys.path.insert(0, str(Path(__file__).resolve().parent))
+        from prometheus_client import CollectorRegistry
+
+        from wva_amd.api import v1alpha1
+        from wva_amd.controller import metrics as ctrl_metrics
+        from wva_amd.controller.promclient import MockPromAPI
+        from wva_amd.controller.reconciler import VariantAutoscalingReconciler
+        from kube_fixtures import (
+            PREMIUM_YAML,
+            make_cluster,
+            make_deployment,
+            make_va,
+            set_load_metrics,
+        )
+        from wva_amd.controller.reconciler import (
+            CONFIG_MAP_NAMESPACE,
+            SERVICE_CLASSES_CM,
+        )
+        from wva_amd.kube import ConfigMap
+
+        def desired_for(analyzer_env):
+            if analyzer_env is None:
+                monkeypatch.delenv("WVA_ANALYZER", raising=False)
+            else:
+                monkeypatch.setenv("WVA_ANALYZER", "mg1")
+                monkeypatch.setenv("WVA_SERVICE_SCV", analyzer_env)
+            cluster = make_cluster()
+            # tighten the Premium TTFT target so waiting time binds
+            cm = cluster.get(ConfigMap, SERVICE_CLASSES_CM, CONFIG_MAP_NAMESPACE)
+            cm.data["premium.yaml"] = PREMIUM_YAML.replace(
+                "slo-ttft: 500", "slo-ttft: 15"
+            ).replace("slo-tpot: 24", "slo-tpot: 0")
+            cluster.update(cm)
+            make_deployment(cluster, replicas=1)
+            make_va(cluster, max_batch=16, alpha="12.0", beta="6.0",
+                    gamma="4.0", delta="0.01")
+            prom = MockPromAPI()
+            set_load_metrics(prom, "default/llama-8b", "default",
+                             arrival_rps=40.0, in_tokens=32.0, out_tokens=25.0)
+            registry = CollectorRegistry()
+            ctrl_metrics.init_metrics(registry)
+            try:
+                VariantAutoscalingReconciler(cluster, prom).reconcile()
+            finally:
+                ctrl_metrics.reset_metrics()
+            va = cluster.get(v1alpha1.VariantAutoscaling, "vllm-llama", "default")
+            return va.status.desired_optimized_alloc.num_replicas
+
+        markovian = desired_for(None)
+        md1 = desired_for("0")
+        assert markovian >= 1 and md1 >= 1
+        assert md1 <= markovian
+        assert md1 < markovian, (
+            f"cs^2=0 should size fewer replicas on a wait-bound workload "
+            f"(markovian={markovian}, md1={md1})"
+        )
