@@ -103,19 +103,6 @@ struct WgEval {
     return b - 1;
   }
 
-  __device__ void window_wave(double loglam, int n_star, double m, int lane,
-                              int *lo_out, int *hi_out) const {
-    const double thresh = m - kLogCutoff;
-    *lo_out = wave_lower_bound(
-        0, n_star, lane, [&](int i) { return log_p(cum, loglam, i) >= thresh; });
-    // right edge: first n in (n_star, K+1] below threshold, minus one
-    // (probes past K count as below, so pred(K+1) is true by definition)
-    const int first_below = wave_lower_bound(
-        n_star, K + 1, lane,
-        [&](int i) { return i > K || log_p(cum, loglam, i) < thresh; });
-    *hi_out = first_below - 1;
-  }
-
   __device__ Stats eval(double lam) const {
     const int tid = threadIdx.x;
     const int wave = tid >> 6;
@@ -124,23 +111,24 @@ struct WgEval {
     (void)lane;
     const double loglam = log(lam);
 
-    // max of logp via the concavity closed form, and the significant
-    // state window via two boundary searches — wave-parallel (ballot)
-    // in the single-wave geometry, scalar per-thread otherwise; the
-    // exp sweep then shrinks to the ~window where p(n) > 1e-16
-    int n_star, n_lo, n_hi;
+    // max of logp via the concavity closed form — wave-parallel
+    // (ballot) in the single-wave geometry, scalar otherwise.  No
+    // windowing: the pre-batch region is at most N+1 states (sub-
+    // threshold terms just underflow toward 0 in the sweep, adding
+    // accuracy, not cost) and the whole queue region [N+1, K] sums in
+    // closed form below — so the two window boundary searches the
+    // windowed design needed are gone entirely.
+    int n_star;
     double m;
     if constexpr (WAVES == 1) {
       n_star = mode_state_wave(lam, lane);
-      m = log_p(cum, loglam, n_star);
-      if (m < 0.0) m = 0.0;
-      window_wave(loglam, n_star, m, lane, &n_lo, &n_hi);
     } else {
       n_star = log_mode_state(p, K, lam);
-      m = log_p(cum, loglam, n_star);
-      if (m < 0.0) m = 0.0;
-      state_window(cum, loglam, K, n_star, m, &n_lo, &n_hi);
     }
+    m = log_p(cum, loglam, n_star);
+    if (m < 0.0) m = 0.0;
+    const int n_lo = 0;
+    const int n_hi = K;
     if constexpr (WAVES > 1) {
       __syncthreads();  // red may still be read from a previous eval
     }
