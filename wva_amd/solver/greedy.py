@@ -86,9 +86,12 @@ def solve_greedy(system: System, spec: OptimizerSpec) -> None:
             e.delta = math.inf
         entries.append(e)
 
-    import functools
-
-    entries.sort(key=functools.cmp_to_key(_order_cmp))
+    # same ordering as _order_cmp — (priority asc, regret delta desc,
+    # current value desc) — as a key tuple: ~10x cheaper than the
+    # cmp_to_key wrapper, and equally stable for ties
+    entries.sort(
+        key=lambda e: (e.priority, -e.delta, -e.allocations[e.cur_index].value)
+    )
 
     if spec.delayed_best_effort:
         unallocated = _allocate(system, entries, available)
