@@ -233,6 +233,46 @@ def collect_gpu_telemetry(prom: PromAPI, namespace: str) -> Optional[GpuTelemetr
 
 
 def collect_inventory_k8s(client) -> Dict[str, Dict[str, object]]:
-    """Limited-mode inventory stub (collector.go:37-42 keeps the same stub);
-    unlimited mode needs no cluster inventory."""
-    return {}
+    """Live GPU inventory from the cluster's Nodes — a working
+    implementation of what the reference stubs out (collector.go:37-42).
+
+    Per node, for each vendor prefix (AMD first), reads the extended-
+    resource count ``<vendor>/gpu`` from status.allocatable (falling
+    back to status.capacity) and the product from the
+    ``<vendor>/gpu.product`` label — the convention installed by
+    deploy/kind-emulator/setup.sh and the AMD GPU operator.  Returns
+    ``{product: {"count": units, "nodes": n, "vendor": prefix}}``;
+    limited mode turns this into the solver's capacity pool (the
+    product name matches the accelerator entry's name in the unit-cost
+    ConfigMap, e.g. "MI355X")."""
+    from ..kube import Node
+
+    out: Dict[str, Dict[str, object]] = {}
+    try:
+        nodes = client.list(Node)
+    except Exception as e:
+        log.warn("node inventory unavailable", error=str(e))
+        return {}
+    for node in nodes:
+        for vendor in VENDORS:
+            raw = node.status.allocatable.get(f"{vendor}/gpu") or node.status.capacity.get(
+                f"{vendor}/gpu"
+            )
+            if not raw:
+                continue
+            try:
+                count = int(raw)
+            except ValueError:
+                log.warn("unparseable gpu capacity on node", node=node.name, value=raw)
+                continue
+            if count <= 0:
+                continue
+            product = node.metadata.labels.get(f"{vendor}/gpu.product", "")
+            if not product:
+                log.warn("node has gpus but no product label; skipping",
+                         node=node.name, vendor=vendor)
+                continue
+            entry = out.setdefault(product, {"count": 0, "nodes": 0, "vendor": vendor})
+            entry["count"] += count
+            entry["nodes"] += 1
+    return out

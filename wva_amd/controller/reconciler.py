@@ -205,7 +205,26 @@ class VariantAutoscalingReconciler:
             return ReconcileResult(None)
 
         mark_phase("config")
-        system_data = create_system_data(accelerator_cm, service_class_cm, optimization_cm)
+        # live GPU inventory: in limited mode with WVA_INVENTORY=k8s the
+        # capacity pool comes from the cluster's Nodes (amd.com/gpu
+        # allocatable + product labels) instead of the static CM field —
+        # a working version of the inventory the reference stubs
+        # (collector.go:37-42); inventory failures degrade to the static
+        # pool, never fail the cycle
+        inventory = None
+        if (
+            optimization_cm.get("WVA_OPTIMIZER_MODE", "").lower() == "limited"
+            and optimization_cm.get("WVA_INVENTORY", "").lower() == "k8s"
+        ):
+            inventory = collector.collect_inventory_k8s(self.client)
+            if inventory:
+                log.info(
+                    "node inventory collected",
+                    products={k: v["count"] for k, v in inventory.items()},
+                )
+        system_data = create_system_data(
+            accelerator_cm, service_class_cm, optimization_cm, inventory=inventory
+        )
         update_list, va_map, responses = self._prepare_variant_autoscalings(
             active, accelerator_cm, service_class_cm, system_data
         )

@@ -151,6 +151,7 @@ def create_system_data(
     accelerator_cm: Dict[str, Dict[str, str]],
     service_class_cm: Dict[str, str],
     optimizer_cm: Optional[Dict[str, str]] = None,
+    inventory: Optional[Dict[str, Dict[str, object]]] = None,
 ) -> SystemData:
     """Adapter from ConfigMap payloads to the optimizer's SystemSpec.
 
@@ -158,10 +159,14 @@ def create_system_data(
     capacity-constrained solver dormant (utils.go:170-173).  Here limited
     mode is reachable: set ``WVA_OPTIMIZER_MODE: limited`` in the
     controller ConfigMap (plus optional ``WVA_SATURATION_POLICY`` and
-    ``WVA_DELAYED_BEST_EFFORT``) and give accelerator entries a
-    ``capacity`` field (units of that type available in the pool)."""
+    ``WVA_DELAYED_BEST_EFFORT``) and give the pool either statically —
+    a ``capacity`` field per accelerator entry — or live, from the
+    cluster's Node inventory (``WVA_INVENTORY: k8s``; ``inventory`` is
+    collector.collect_inventory_k8s output keyed by product name, which
+    takes precedence over the static field)."""
     sd = SystemData(spec=SystemSpec())
     optimizer_cm = optimizer_cm or {}
+    inventory = inventory or {}
 
     for key, val in accelerator_cm.items():
         try:
@@ -182,7 +187,12 @@ def create_system_data(
         except ValueError:
             pass
         sd.spec.accelerators.spec.append(spec)
-        if "capacity" in val:
+        if key in inventory:
+            # live node inventory wins over the static capacity field
+            sd.spec.capacity.count.append(
+                AcceleratorCount(type=spec.type, count=int(inventory[key]["count"]))
+            )
+        elif "capacity" in val:
             try:
                 sd.spec.capacity.count.append(
                     AcceleratorCount(type=spec.type, count=int(val["capacity"]))

@@ -14,7 +14,7 @@ ownerReference garbage collection, resourceVersion bumping) behind a small
 """
 
 from .errors import ConflictError, ForbiddenError, GoneError, InvalidError, KubeError, NotFoundError
-from .objects import ConfigMap, Deployment, DeploymentSpec, DeploymentStatus, Lease, LeaseSpec
+from .objects import ConfigMap, Deployment, DeploymentSpec, DeploymentStatus, Lease, LeaseSpec, Node, NodeStatus
 from .client import InMemoryKubeClient, KubeClient
 from .http_client import HTTPKubeClient
 
@@ -29,6 +29,8 @@ __all__ = [
     "Deployment",
     "Lease",
     "LeaseSpec",
+    "Node",
+    "NodeStatus",
     "DeploymentSpec",
     "DeploymentStatus",
     "KubeClient",

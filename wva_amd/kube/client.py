@@ -28,7 +28,7 @@ from typing import Callable, Dict, List, Optional, Protocol, Tuple, Type, TypeVa
 
 from ..api.v1alpha1.types import VariantAutoscaling
 from .errors import ConflictError, GoneError, NotFoundError
-from .objects import ConfigMap, Deployment, Lease
+from .objects import ConfigMap, Deployment, Lease, Node
 
 T = TypeVar("T")
 
@@ -37,6 +37,7 @@ _KINDS = {
     ConfigMap: "ConfigMap",
     Deployment: "Deployment",
     Lease: "Lease",
+    Node: "Node",  # cluster-scoped: stored under namespace ""
 }
 
 # Watch history bound: events older than the newest EVENT_LOG_LIMIT are

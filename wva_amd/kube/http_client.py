@@ -38,17 +38,19 @@ from .errors import (
     KubeError,
     NotFoundError,
 )
-from .objects import ConfigMap, Deployment, Lease
+from .objects import ConfigMap, Deployment, Lease, Node
 
 T = TypeVar("T")
 
 SA_DIR = "/var/run/secrets/kubernetes.io/serviceaccount"
 
 _RESOURCES = {
+    # (prefix, plural, namespaced)
     VariantAutoscaling: ("apis/llmd.ai/v1alpha1", "variantautoscalings", True),
-    ConfigMap: ("api/v1", "configmaps", False),
+    ConfigMap: ("api/v1", "configmaps", True),
     Deployment: ("apis/apps/v1", "deployments", True),
     Lease: ("apis/coordination.k8s.io/v1", "leases", True),
+    Node: ("api/v1", "nodes", False),
 }
 
 
@@ -110,8 +112,11 @@ class HTTPKubeClient:
             raise TypeError(f"unregistered kind {cls.__name__}") from None
 
     def _path(self, cls_or_obj, namespace: str, name: str = "", subresource: str = "") -> str:
-        prefix, plural, _ = self._resource(cls_or_obj)
-        path = f"/{prefix}/namespaces/{namespace}/{plural}"
+        prefix, plural, namespaced = self._resource(cls_or_obj)
+        if namespaced:
+            path = f"/{prefix}/namespaces/{namespace}/{plural}"
+        else:  # cluster-scoped (e.g. Node)
+            path = f"/{prefix}/{plural}"
         if name:
             path += f"/{name}"
         if subresource:

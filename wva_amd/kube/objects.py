@@ -80,3 +80,30 @@ class Deployment(_Base):
     @property
     def namespace(self) -> str:
         return self.metadata.namespace
+
+
+class NodeStatus(_Base):
+    capacity: Dict[str, str] = Field(default_factory=dict)
+    allocatable: Dict[str, str] = Field(default_factory=dict)
+
+
+class Node(_Base):
+    """core/v1 Node (cluster-scoped) — carries the GPU inventory the
+    limited-mode solver consumes: ``amd.com/gpu`` extended-resource
+    counts in status.allocatable plus the ``amd.com/gpu.product`` label
+    (the convention deploy/kind-emulator/setup.sh and
+    deploy/emulator/amd-gpu-node-labels.yaml install).  The reference
+    stubs this inventory (collector.go:37-42); here it is live."""
+
+    api_version: str = Field(alias="apiVersion", default="v1")
+    kind: str = "Node"
+    metadata: ObjectMeta = Field(default_factory=ObjectMeta)
+    status: NodeStatus = Field(default_factory=NodeStatus)
+
+    @property
+    def name(self) -> str:
+        return self.metadata.name
+
+    @property
+    def namespace(self) -> str:
+        return self.metadata.namespace

@@ -34,7 +34,7 @@ from fastapi.responses import JSONResponse, StreamingResponse
 
 from .client import InMemoryKubeClient
 from .errors import ConflictError, GoneError, NotFoundError
-from .objects import ConfigMap, Deployment, Lease
+from .objects import ConfigMap, Deployment, Lease, Node
 from .schema import CRDValidator, SchemaValidationError
 
 _ROUTES: Dict[str, Type] = {
@@ -286,5 +286,46 @@ def create_stub_api_server(store: Optional[InMemoryKubeClient] = None):
                 return handle(lambda: store.delete(cls, name, namespace) or {"status": "Success"})
 
         make_routes()
+
+    # cluster-scoped core/v1 Nodes (GPU inventory for limited mode)
+    @app.get("/api/v1/nodes")
+    async def list_nodes(
+        watch: bool = False,
+        timeoutSeconds: int = 30,
+        resourceVersion: str = "",
+        allowWatchBookmarks: bool = False,
+        limit: int = 0,
+        request: Request = None,
+    ):
+        if watch:
+            return watch_stream(
+                Node, None, timeoutSeconds, resourceVersion, allowWatchBookmarks
+            )
+        cont = request.query_params.get("continue", "") if request else ""
+        return list_response(Node, None, limit, cont)
+
+    @app.get("/api/v1/nodes/{name}")
+    async def get_node(name: str):
+        return handle(lambda: dump(store.get(Node, name, "")))
+
+    @app.post("/api/v1/nodes")
+    async def create_node(request: Request):
+        body = await request.json()
+
+        def do():
+            obj = Node.model_validate(body)
+            obj.metadata.namespace = ""
+            return dump(store.create(obj))
+
+        return handle(do)
+
+    @app.put("/api/v1/nodes/{name}")
+    async def update_node(name: str, request: Request):
+        body = await request.json()
+        return handle(lambda: dump(store.update(Node.model_validate(body))))
+
+    @app.delete("/api/v1/nodes/{name}")
+    async def delete_node(name: str):
+        return handle(lambda: store.delete(Node, name, "") or {"status": "Success"})
 
     return app, store
