@@ -393,3 +393,94 @@ class TestKindEmulatorAssets:
         vals = yaml.safe_load((DEPLOY / "emulator" / "prometheus-tls-values.yaml").read_text())
         tls = vals["prometheus"]["prometheusSpec"]["web"]["tlsConfig"]
         assert tls["cert"]["secret"]["name"] == "prometheus-tls"
+
+
+class TestMainInClusterStartup:
+    def test_main_in_cluster_against_stub_apiserver(self, monkeypatch, tmp_path):
+        """The released image's exact path: main() with
+        --kube-backend=in-cluster drives HTTPKubeClient against a (stub)
+        API server and a TLS Prometheus, reconciles once, and writes the
+        VA status through the wire."""
+        import subprocess
+        import threading
+        import time
+
+        import uvicorn
+        from fastapi import FastAPI
+
+        from wva_amd.api import v1alpha1
+        from wva_amd.kube.stub_server import create_stub_api_server
+        import sys as _sys
+        from pathlib import Path as _Path
+
+        _sys.path.insert(0, str(_Path(__file__).resolve().parent))
+        from kube_fixtures import make_cluster, make_deployment, make_va
+
+        # TLS prometheus answering every query with a fresh value
+        crt, key = tmp_path / "tls.crt", tmp_path / "tls.key"
+        subprocess.run(
+            ["openssl", "req", "-x509", "-newkey", "rsa:2048", "-nodes",
+             "-keyout", str(key), "-out", str(crt), "-days", "2",
+             "-subj", "/CN=127.0.0.1", "-addext", "subjectAltName=IP:127.0.0.1"],
+            check=True, capture_output=True,
+        )
+        prom_app = FastAPI()
+
+        @prom_app.get("/api/v1/query")
+        async def q(query: str = ""):
+            return {
+                "status": "success",
+                "data": {"resultType": "vector",
+                         "result": [{"metric": {}, "value": [time.time(), "2"]}]},
+            }
+
+        prom = uvicorn.Server(
+            uvicorn.Config(prom_app, host="127.0.0.1", port=0, log_level="error",
+                           ssl_certfile=str(crt), ssl_keyfile=str(key))
+        )
+        prom_t = threading.Thread(target=prom.run, daemon=True)
+        prom_t.start()
+
+        # stub apiserver seeded with the cluster fixtures + one workload
+        store = make_cluster(opt_interval="1s")
+        make_deployment(store, name="vllm-llama", replicas=1)
+        make_va(store, name="vllm-llama", max_batch=16)
+        app, _ = create_stub_api_server(store)
+        kube = uvicorn.Server(
+            uvicorn.Config(app, host="127.0.0.1", port=0, log_level="error")
+        )
+        kube_t = threading.Thread(target=kube.run, daemon=True)
+        kube_t.start()
+        for s in (prom, kube):
+            for _ in range(200):
+                if s.started:
+                    break
+                time.sleep(0.05)
+        prom_port = prom.servers[0].sockets[0].getsockname()[1]
+        kube_port = kube.servers[0].sockets[0].getsockname()[1]
+        try:
+            monkeypatch.setenv("PROMETHEUS_BASE_URL", f"https://127.0.0.1:{prom_port}")
+            monkeypatch.setenv("PROMETHEUS_CA_CERT_PATH", str(crt))
+            import wva_amd.kube.http_client as http_client_mod
+
+            monkeypatch.setattr(
+                http_client_mod,
+                "in_cluster_config",
+                lambda: {"base_url": f"http://127.0.0.1:{kube_port}",
+                         "token": "sa-token", "ca_cert_path": ""},
+            )
+            from wva_amd.__main__ import main
+
+            rc = main(
+                ["--kube-backend", "in-cluster", "--max-cycles", "1",
+                 "--health-probe-bind-address", ":0"]
+            )
+            assert rc == 0
+            va = store.get(v1alpha1.VariantAutoscaling, "vllm-llama", "default")
+            assert va.status.desired_optimized_alloc.num_replicas >= 1
+            assert v1alpha1.is_condition_true(va, v1alpha1.TYPE_OPTIMIZATION_READY)
+        finally:
+            prom.should_exit = True
+            kube.should_exit = True
+            prom_t.join(timeout=5)
+            kube_t.join(timeout=5)
