@@ -573,3 +573,44 @@ class TestAnalyzerAutoWiring:
         monkeypatch.setenv("WVA_BATCHED_ANALYZER", "1")
         r = VariantAutoscalingReconciler(cluster, prom, batched_analyzer=False)
         assert r.batched_analyzer is False
+
+
+class TestSizingHeadroom:
+    def test_default_is_exact_parity(self, cluster, prom, registry, monkeypatch):
+        monkeypatch.delenv("WVA_SIZING_HEADROOM", raising=False)
+        from wva_amd.core.allocation import sizing_headroom
+
+        assert sizing_headroom() == 0.0
+
+    def test_headroom_inflates_replicas(self, cluster, prom, registry, monkeypatch):
+        """Opt-in WVA_SIZING_HEADROOM sizes for load x (1+h) — the knob
+        that absorbs the ramp-transient misses the endurance soak
+        measured — while status still reports the measured load."""
+        make_deployment(cluster, replicas=1)
+        make_va(cluster)
+        set_load_metrics(prom, "default/llama-8b", "default", arrival_rps=8.0, out_tokens=200.0)
+
+        def desired():
+            rec = VariantAutoscalingReconciler(cluster, prom)
+            rec.reconcile()
+            va = cluster.get(v1alpha1.VariantAutoscaling, "vllm-llama", "default")
+            return (
+                va.status.desired_optimized_alloc.num_replicas,
+                float(va.status.current_alloc.load.arrival_rate),
+            )
+
+        monkeypatch.delenv("WVA_SIZING_HEADROOM", raising=False)
+        base, base_rate = desired()
+        monkeypatch.setenv("WVA_SIZING_HEADROOM", "0.5")
+        inflated, inflated_rate = desired()
+        assert inflated > base
+        # measured load in status is untouched by the sizing headroom
+        assert inflated_rate == base_rate
+
+    def test_garbage_value_ignored(self, monkeypatch):
+        from wva_amd.core.allocation import sizing_headroom
+
+        monkeypatch.setenv("WVA_SIZING_HEADROOM", "lots")
+        assert sizing_headroom() == 0.0
+        monkeypatch.setenv("WVA_SIZING_HEADROOM", "-0.3")
+        assert sizing_headroom() == 0.0

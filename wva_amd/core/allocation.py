@@ -19,6 +19,7 @@ de-facto contract quirks:
 from __future__ import annotations
 
 import math
+import os
 from typing import TYPE_CHECKING, Optional, Tuple
 
 from ..analyzer import (
@@ -131,6 +132,24 @@ class Allocation:
         )
 
 
+def sizing_headroom() -> float:
+    """Opt-in provisioning headroom (WVA_SIZING_HEADROOM, e.g. "0.25"):
+    the fleet is sized for measured load x (1 + h), absorbing ramp
+    transients between reconcile cycles (the endurance soaks' only
+    SLO misses were stages served at the PREVIOUS cycle's size).  The
+    reference has no equivalent; 0 (default) keeps exact parity.
+    Measured load reported in status stays untouched — headroom applies
+    to sizing only."""
+    raw = os.environ.get("WVA_SIZING_HEADROOM", "").strip()
+    if not raw:
+        return 0.0
+    try:
+        h = float(raw)
+    except ValueError:
+        return 0.0
+    return max(h, 0.0)
+
+
 def create_allocation(system: "System", server_name: str, acc_name: str) -> Optional[Allocation]:
     """Create an allocation of accelerator ``acc_name`` to ``server_name``;
     ``None`` if infeasible.  Takes the system explicitly (no singleton)."""
@@ -196,6 +215,7 @@ def create_allocation(system: "System", server_name: str, acc_name: str) -> Opti
         total_rate = load.arrival_rate / 60.0  # req/min -> req/s
     else:
         total_rate = target.tps / float(K)
+    total_rate *= 1.0 + sizing_headroom()
     num_replicas = int(math.ceil(total_rate / rate_star))
     num_replicas = max(num_replicas, server.min_num_replicas)
 

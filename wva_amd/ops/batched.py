@@ -266,6 +266,9 @@ class BatchedAllocationSolver:
             pair_cache[key] = info
             return info
 
+        from ..core.allocation import sizing_headroom
+
+        headroom_factor = 1.0 + sizing_headroom()
         rows: List[Tuple[float, ...]] = []
         keys: List[Tuple[str, str, int]] = []  # (server, acc, N)
         zero_load: Dict[str, Dict[str, Allocation]] = {}
@@ -309,7 +312,9 @@ class BatchedAllocationSolver:
                 alpha, beta, gamma, delta, perf_max_batch, at_tokens, cost_per_rep = info
                 K = int(out_tok)
                 N = server_max_batch if server_max_batch > 0 else max(perf_max_batch * at_tokens // K, 1)
-                total_rate = arrival / 60.0 if target_tps == 0 else target_tps / float(K)
+                total_rate = (
+                    arrival / 60.0 if target_tps == 0 else target_tps / float(K)
+                ) * headroom_factor
                 rows_append(
                     (alpha, beta, gamma, delta, float(int(in_tok)), float(K), float(N),
                      target_ttft, target_itl, target_tps, total_rate, min_replicas)
