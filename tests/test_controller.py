@@ -587,8 +587,9 @@ class TestSizingHeadroom:
         that absorbs the ramp-transient misses the endurance soak
         measured — while status still reports the measured load."""
         make_deployment(cluster, replicas=1)
-        make_va(cluster)
-        set_load_metrics(prom, "default/llama-8b", "default", arrival_rps=8.0, out_tokens=200.0)
+        make_va(cluster, max_batch=16, alpha="12.0", beta="6.0", gamma="4.0", delta="0.01")
+        set_load_metrics(prom, "default/llama-8b", "default", arrival_rps=8.0,
+                         in_tokens=32.0, out_tokens=25.0)
 
         def desired():
             rec = VariantAutoscalingReconciler(cluster, prom)
