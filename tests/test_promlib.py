@@ -189,3 +189,19 @@ class TestScraperLoop:
         n = calls["n"]
         time.sleep(0.15)
         assert calls["n"] == n
+
+
+class TestRangeScanRewrite:
+    def test_inclusive_bounds_and_order(self):
+        # regression for the reversed-scan rewrite: inclusive [start, end],
+        # ascending output, early exit must not drop edge samples
+        from wva_amd.promlib.store import TimeSeriesStore
+
+        st = TimeSeriesStore(retention=1000)
+        for i in range(10):
+            st.add_sample("m", {}, float(i), ts=100.0 + i)
+        s = st.select("m", {})[0]
+        assert s.range(102.0, 105.0) == [(102.0, 2.0), (103.0, 3.0), (104.0, 4.0), (105.0, 5.0)]
+        assert s.range(108.5, 200.0) == [(109.0, 9.0)]
+        assert s.range(0.0, 99.0) == []
+        assert s.range(109.0, 109.0) == [(109.0, 9.0)]
