@@ -521,11 +521,25 @@ __device__ inline void mb_publish(volatile double *slot, double y,
   *flag = round_id;
 }
 
-__device__ inline void mb_wait(volatile double *flags, int n, double round_id) {
+// Watchdog: a legitimate wait is bounded by a sibling wave's evaluation
+// (microseconds); ~16M sleep iterations is on the order of a second of
+// GPU time.  If a (hypothetical) control-flow divergence bug ever broke
+// the uniform-rounds invariant, the spin gives up instead of hanging
+// the device — every wave of the group times out the same way, all
+// fall through, and the problem's zeroed result row reads as
+// infeasible: wrong-but-finite beats a dead GPU.
+constexpr unsigned kMailboxSpinLimit = 1u << 24;
+
+__device__ inline bool mb_wait(volatile double *flags, int n, double round_id) {
   for (int j = 0; j < n; ++j) {
-    while (flags[j] < round_id) __builtin_amdgcn_s_sleep(2);
+    unsigned spins = 0;
+    while (flags[j] < round_id) {
+      __builtin_amdgcn_s_sleep(2);
+      if (++spins > kMailboxSpinLimit) return false;
+    }
   }
   __threadfence_block();
+  return true;
 }
 
 // Midpoint at heap node k (1-based: 1 = root midpoint, 2/3 = children,
@@ -563,7 +577,10 @@ __device__ int spec_binary_search(double x_min, double x_max, double y_target,
   if (sub == 0) mine = eval(x_min);
   if (sub == 1) mine = eval(x_max);
   if (lane == 0 && sub < 2) mb_publish(&yb[sub], mine, &fl[sub], 1.0);
-  mb_wait(fl, 2, 1.0);
+  if (!mb_wait(fl, 2, 1.0)) {  // watchdog: treat as unreachable target
+    *x_star = x_min;
+    return -1;
+  }
   const double y0 = yb[0];
   const double y1 = yb[1];
 
@@ -601,7 +618,10 @@ __device__ int spec_binary_search(double x_min, double x_max, double y_target,
     const double x_mine = heap_midpoint(lo, hi, sub + 1);
     const double y_mine = eval(x_mine);
     if (lane == 0) mb_publish(&yb[sub], y_mine, &fl[sub], round_id);
-    mb_wait(fl, SW, round_id);
+    if (!mb_wait(fl, SW, round_id)) {
+      *x_star = x_min;
+      return -1;
+    }
     round_id += 1.0;
 
     // replay the sequential walk on the cached tree
@@ -695,7 +715,7 @@ __device__ void solve_body_spec(const double *__restrict__ prob,
   }
   if (wave != 0) return;  // group-0 wave 0 finishes the tail alone
 
-  mb_wait(done, 2, 1.0);
+  if (!mb_wait(done, 2, 1.0)) return;  // watchdog: leave result infeasible
   if (ex[1] < 0.0 || ex[3] < 0.0) return;  // a target below reachable range
 
   double lam_tps = lam_max;
