@@ -1,31 +1,35 @@
 // gfx950 (CDNA4) batched allocation-sizing kernel.
 //
-// One 256-thread workgroup (4 wave64) per problem.  The whole analytic
-// pipeline of core.create_allocation (service-rate table, log-space
-// product-form M/M/1/K probabilities, TTFT/ITL bisection, replica count,
-// final per-replica analysis) runs in-workgroup:
+// One workgroup per problem: the whole analytic pipeline of
+// core.create_allocation (service-rate table, log-space product-form
+// M/M/1/K probabilities, TTFT/ITL bisection, replica count, final
+// per-replica analysis) runs in-workgroup:
 //
-//   - the cumulative log-service-rate table (K = 11*N doubles) lives in LDS
-//     and is built with a chunked parallel scan — every later model
-//     evaluation is LDS-bandwidth bound, never HBM;
-//   - each model evaluation is ONE strided pass over the significant
-//     state window (the normalization max and the ~1e-16 cutoff window
-//     both come from closed-form O(log K) searches over the concave
-//     log-probability curve) with wave64 __shfl_down reductions;
-//     consecutive lanes touch consecutive doubles -> conflict-free
-//     ds_read_b64 (bank = (a/4) % 64);
-//   - the bisection control flow is uniform across the workgroup (all
-//     decisions derive from broadcast reduction results), so the
-//     __syncthreads() inside evaluations are safe;
-//   - LDS budget: (K + 256 + 32) doubles <= 64 KiB for N <= ~700 (the
-//     binding routes larger batch limits to the CPU path).  At N = 256
-//     (the collector's default max batch) that is ~25 KiB -> 2+ workgroups
-//     per CU, and a fleet-sized batch (hundreds of (server, accelerator)
-//     pairs) fills all 256 CUs.
+//   - the cumulative log-service-rate table (K = 11*N doubles) lives in
+//     LDS and is built with a chunked parallel scan — every later model
+//     evaluation is LDS-bound, never HBM; consecutive lanes touch
+//     consecutive doubles -> conflict-free ds_read_b64;
+//   - a model evaluation sweeps only the <= N+1 pre-batch states (wave64
+//     __shfl_down reductions, 2x-unrolled exp); the whole constant-rate
+//     queue region [N+1, K] — 10/11 of the states — sums in CLOSED FORM
+//     (geo_tail in queue_core.h, expm1-based geometric sums), which also
+//     made the 1e-16 significance window obsolete;
+//   - the normalization mode comes from a wave-parallel 64-probe ballot
+//     search (wave_lower_bound) instead of a serial bisection;
+//   - the default spec geometry (384 = 2 search groups x 3 wave64) runs
+//     the TTFT and ITL bisections concurrently AND speculatively: each
+//     round a group evaluates its bracket's full depth-2 midpoint tree
+//     (bitwise the sequential iterates) and consumes two bisection
+//     levels; wave sync is group-local LDS mailboxes with a watchdog —
+//     no workgroup barriers after the scan;
+//   - LDS budget: (max_k + THREADS + 64) doubles <= 64 KiB for N <= ~700
+//     (the binding routes larger batch limits to the CPU path).
 //
-// Numerics are double throughout, matching the Python analyzer; math
-// parity is covered by tests/test_ops.py (CPU) and the @gpu numerics
-// tests.
+// Numerics are double throughout, matching the Python analyzer; parity
+// is enforced by tests/test_ops.py, the @gpu numerics tests and
+// tools/fuzz_parity.py (20k problems, zero feasibility flips).  Design
+// narrative + measurements: docs/design/native-queue-solver.md,
+// profiles/r02_kernel_notes.md.
 
 #include <hip/hip_runtime.h>
 
@@ -33,19 +37,16 @@
 
 namespace wva {
 
-// Three workgroup geometries are instantiated:
-//   - 256 threads (4 wave64): strided sweeps + LDS cross-wave combine;
-//   - 64 threads (1 wave64): barrier-free — the wave executes in lockstep,
-//     reductions are pure __shfl_down chains and the combined values are
-//     broadcast from lane 0 with __shfl (no LDS round-trip);
-//   - 128 threads (2 wave64, "dual"): wave 0 runs the TTFT bisection
-//     while wave 1 concurrently runs the ITL bisection, each with its own
-//     barrier-free 64-lane evaluations — for the common both-targets
-//     case this halves the serial-bisection critical path, which is what
-//     bounds small-fleet dispatches (a 192-problem launch cannot fill
-//     256 CUs, so per-problem latency IS the dispatch time).
-// The launcher auto-selects by state-chain length; the WVA_GPU_THREADS
-// env var (64|128|256) overrides, read per launch.
+// Five workgroup geometries are instantiated:
+//   - 64 (sequential single wave, barrier-free __shfl reductions);
+//   - 128 ("dual": wave 0 searches TTFT while wave 1 searches ITL —
+//     the winner for chip-filling launches, B >= 512);
+//   - 256 (4-wave strided sweeps + LDS cross-wave combine);
+//   - 384 (speculative-tree multisection, depth 2 — the default for
+//     underfilled, latency-bound launches);
+//   - 896 (depth-3 speculation; measured slower than depth 2).
+// The launcher auto-selects (occupancy-aware, see wva_launch_solve);
+// WVA_GPU_THREADS=64|128|256|384|896 overrides, read per launch.
 
 // Reduction scratch layout (doubles, after cum[max_k] in dynamic LDS):
 //   red[0..WAVES*5-1]  per-wave partials (S, Ni, Snum, Ninum, eK)
