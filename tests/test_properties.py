@@ -152,3 +152,77 @@ class TestGreedyProperties:
                 continue
             best = min(a.value for a in server.all_allocations.values())
             assert server.allocation.value == best
+
+
+class TestGreedyPriorityProperties:
+    @settings(max_examples=15, deadline=None)
+    @given(
+        st.integers(1, 4),   # premium servers
+        st.integers(1, 4),   # freemium servers
+        st.integers(1, 6),   # capacity units
+    )
+    def test_priority_prefix_under_scarcity(self, n_prem, n_free, cap):
+        """With identical server shapes on a single accelerator type and
+        the None policy, the allocated set must be a priority-ordered
+        prefix: no Freemium server holds capacity while a Premium server
+        went unallocated."""
+        from wva_amd.solver import Solver
+
+        servers = [
+            server_spec(f"p{i}:ns", class_name="Premium", arrival_rate=600.0)
+            for i in range(n_prem)
+        ] + [
+            server_spec(f"f{i}:ns", class_name="Freemium", arrival_rate=600.0)
+            for i in range(n_free)
+        ]
+        system, opt = make_system(
+            servers=servers,
+            unlimited=False,
+            capacity=[("AMD-MI355X-288GB", cap)],
+            saturation_policy="None",
+        )
+        system.remove_accelerator("L40S")
+        system.remove_accelerator("MI300X")
+        system.calculate()
+        Solver(opt).solve(system)
+        prem_unallocated = any(
+            system.servers[f"p{i}:ns"].allocation is None for i in range(n_prem)
+        )
+        free_allocated = any(
+            system.servers[f"f{i}:ns"].allocation is not None for i in range(n_free)
+        )
+        if prem_unallocated:
+            # identical shapes: anything a Freemium got, the missing
+            # Premium could have used
+            assert not free_allocated
+
+    @settings(max_examples=10, deadline=None)
+    @given(
+        st.floats(0.0, 2.0),
+        st.floats(60.0, 6000.0),
+    )
+    def test_headroom_monotone_in_h(self, h, rate):
+        """Sized replicas are non-decreasing in WVA_SIZING_HEADROOM."""
+        import os
+
+        from wva_amd.core.allocation import create_allocation
+
+        def replicas(headroom):
+            old = os.environ.get("WVA_SIZING_HEADROOM")
+            os.environ["WVA_SIZING_HEADROOM"] = str(headroom)
+            try:
+                system, _ = make_system(
+                    servers=[server_spec("s:ns", arrival_rate=rate)]
+                )
+                alloc = create_allocation(system, "s:ns", "MI355X")
+                return alloc.num_replicas if alloc else None
+            finally:
+                if old is None:
+                    os.environ.pop("WVA_SIZING_HEADROOM", None)
+                else:
+                    os.environ["WVA_SIZING_HEADROOM"] = old
+
+        base = replicas(0.0)
+        more = replicas(h)
+        if base is not None and more is not None:
+            assert more >= base
