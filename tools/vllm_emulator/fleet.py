@@ -78,7 +78,9 @@ class EmulatorFleet:
         import sys
         from pathlib import Path
 
-        sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
+        tools_dir = str(Path(__file__).resolve().parent.parent)
+        if tools_dir not in sys.path:
+            sys.path.insert(0, tools_dir)
         from loadgen import PoissonLoadGenerator, Stage
 
         active = self.active_urls
