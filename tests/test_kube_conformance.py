@@ -485,3 +485,40 @@ class TestStatusConflictConcurrentWriters:
             assert final.status.desired_optimized_alloc.num_replicas == 7
         finally:
             client.delete(v1alpha1.VariantAutoscaling, "race-helper", "default")
+
+
+class TestSeenUidCompaction:
+    def test_relist_compacts_seen_uids(self, api):
+        """Past SEEN_UIDS_LIMIT the session forces a compacting re-list,
+        so deleted objects' uids don't accumulate forever."""
+        client, store = api
+        stop = threading.Event()
+        session = CreateWatchSession(
+            client,
+            v1alpha1.VariantAutoscaling,
+            namespace="compact",
+            window_seconds=1,
+            stop_event=stop,
+        )
+        session.SEEN_UIDS_LIMIT = 4  # tiny, to reach the branch
+        got = []
+        t = threading.Thread(target=session.run, args=(got.append,), daemon=True)
+        t.start()
+        try:
+            # churn: create+delete past the limit
+            for i in range(8):
+                make_va(store, name=f"churny-{i}", namespace="compact")
+            deadline = time.monotonic() + 8
+            while len(got) < 8 and time.monotonic() < deadline:
+                time.sleep(0.05)
+            assert len(got) == 8
+            for i in range(8):
+                store.delete(v1alpha1.VariantAutoscaling, f"churny-{i}", "compact")
+            # after a window completes, a compacting re-list empties the set
+            deadline = time.monotonic() + 8
+            while len(session._seen_uids) > 0 and time.monotonic() < deadline:
+                time.sleep(0.1)
+            assert len(session._seen_uids) == 0
+        finally:
+            stop.set()
+            t.join(timeout=8)

@@ -373,7 +373,7 @@ class CreateWatchSession:
                         self._seen_uids.add(obj.metadata.uid)
                         callback(obj)
                 self.backoff_s = self.BACKOFF_BASE_S
-                if len(self._seen_uids) > 65536:
+                if len(self._seen_uids) > self.SEEN_UIDS_LIMIT:
                     primed = False  # force a compacting re-list
             except GoneError:
                 # watch history compacted: re-list immediately, resume
