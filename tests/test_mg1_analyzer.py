@@ -244,3 +244,140 @@ class TestMG1ThroughReconciler:
             f"cs^2=0 should size fewer replicas on a wait-bound workload "
             f"(markovian={markovian}, md1={md1})"
         )
+
+
+class TestAutoScvEstimation:
+    def _bucket_samples(self, cum_by_le):
+        import time as _time
+
+        from wva_amd.controller.promclient import Sample
+
+        now = _time.time()
+        return [
+            Sample(value=v, timestamp=now, labels={"le": le})
+            for le, v in cum_by_le.items()
+        ]
+
+    def test_estimator_concentrated_vs_spread(self):
+        from wva_amd.controller import collector
+        from wva_amd.controller.promclient import MockPromAPI
+
+        prom = MockPromAPI()
+        q = collector.token_scv_query("m", "ns")
+        # concentrated: nearly all mass in one bucket -> tiny scv
+        prom.query_results[q] = self._bucket_samples(
+            {"16": 0.0, "32": 9.9, "64": 10.0, "+Inf": 10.0}
+        )
+        tight = collector.estimate_token_scv(prom, "m", "ns")
+        # spread: mass across 3 octaves -> large scv
+        prom.query_results[q] = self._bucket_samples(
+            {"16": 4.0, "64": 7.0, "512": 9.0, "+Inf": 10.0}
+        )
+        wide = collector.estimate_token_scv(prom, "m", "ns")
+        assert tight is not None and wide is not None
+        assert 0.0 <= tight < 0.2
+        assert wide > tight * 3
+
+    def test_estimator_absent_histogram_returns_none(self):
+        from wva_amd.controller import collector
+        from wva_amd.controller.promclient import MockPromAPI
+
+        prom = MockPromAPI()
+        q = collector.token_scv_query("m", "ns")
+        prom.query_results[q] = []
+        # fallback (namespace-less) also empty
+        prom.query_results[
+            f'rate(vllm:request_generation_tokens_bucket{{model_name="m"}}'
+            f"[{collector.rate_window()}])"
+        ] = []
+        assert collector.estimate_token_scv(prom, "m", "ns") is None
+
+    def test_recommended_scv_damped_by_fixed_time(self):
+        from wva_amd.analyzer import recommended_service_scv
+
+        # long outputs, tiny fixed part -> near the token scv
+        near = recommended_service_scv(200.0, 0.8, alpha=10.0, beta=0.1,
+                                       gamma=1.0, delta=0.0, in_tokens=0)
+        # short outputs, big prefill -> strongly damped
+        damped = recommended_service_scv(5.0, 0.8, alpha=10.0, beta=0.1,
+                                         gamma=500.0, delta=0.0, in_tokens=0)
+        assert 0.5 < near <= 1.0
+        assert damped < near / 4
+
+    def test_auto_mode_end_to_end_through_reconciler(self, monkeypatch):
+        """WVA_SERVICE_SCV=auto: a concentrated token histogram drives a
+        low per-server cs^2 and fewer replicas than the Markovian
+        default on a wait-bound workload."""
+        import sys
+        from pathlib import Path
+
+        sys.path.insert(0, str(Path(__file__).resolve().parent))
+        from prometheus_client import CollectorRegistry
+
+        from wva_amd.api import v1alpha1
+        from wva_amd.controller import collector
+        from wva_amd.controller import metrics as ctrl_metrics
+        from wva_amd.controller.promclient import MockPromAPI
+        from wva_amd.controller.reconciler import (
+            CONFIG_MAP_NAMESPACE,
+            SERVICE_CLASSES_CM,
+            VariantAutoscalingReconciler,
+        )
+        from wva_amd.kube import ConfigMap
+        from kube_fixtures import (
+            PREMIUM_YAML,
+            make_cluster,
+            make_deployment,
+            make_va,
+            set_load_metrics,
+        )
+
+        def desired(env_scv):
+            if env_scv is None:
+                monkeypatch.delenv("WVA_ANALYZER", raising=False)
+                monkeypatch.delenv("WVA_SERVICE_SCV", raising=False)
+            else:
+                monkeypatch.setenv("WVA_ANALYZER", "mg1")
+                monkeypatch.setenv("WVA_SERVICE_SCV", env_scv)
+            cluster = make_cluster()
+            cm = cluster.get(ConfigMap, SERVICE_CLASSES_CM, CONFIG_MAP_NAMESPACE)
+            cm.data["premium.yaml"] = PREMIUM_YAML.replace(
+                "slo-ttft: 500", "slo-ttft: 15"
+            ).replace("slo-tpot: 24", "slo-tpot: 0")
+            cluster.update(cm)
+            make_deployment(cluster, replicas=1)
+            make_va(cluster, max_batch=16, alpha="12.0", beta="6.0",
+                    gamma="4.0", delta="0.01")
+            prom = MockPromAPI()
+            set_load_metrics(prom, "default/llama-8b", "default",
+                             arrival_rps=40.0, in_tokens=32.0, out_tokens=25.0)
+            # concentrated histogram: all requests 24-26 tokens
+            import time as _time
+
+            from wva_amd.controller.promclient import Sample
+
+            now = _time.time()
+            prom.query_results[
+                collector.token_scv_query("default/llama-8b", "default")
+            ] = [
+                Sample(value=0.0, timestamp=now, labels={"le": "24"}),
+                Sample(value=10.0, timestamp=now, labels={"le": "26"}),
+                Sample(value=10.0, timestamp=now, labels={"le": "+Inf"}),
+            ]
+            registry = CollectorRegistry()
+            ctrl_metrics.init_metrics(registry)
+            try:
+                VariantAutoscalingReconciler(cluster, prom).reconcile()
+            finally:
+                ctrl_metrics.reset_metrics()
+            va = cluster.get(v1alpha1.VariantAutoscaling, "vllm-llama", "default")
+            return va.status.desired_optimized_alloc.num_replicas
+
+        markovian = desired(None)
+        auto = desired("auto")
+        assert markovian >= 1 and auto >= 1
+        assert auto < markovian, (
+            f"auto cs^2 from a concentrated histogram must size fewer "
+            f"replicas than mm1k on a wait-bound workload "
+            f"(mm1k={markovian}, auto={auto})"
+        )

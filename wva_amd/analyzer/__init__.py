@@ -10,7 +10,7 @@ is thread-safe by construction).
 
 from .queuemodel import QueueModel, MM1KModel
 from .statedep import MM1ModelStateDependent
-from .mg1 import MG1Corrector, MG1Metrics, configured_scv, pollaczek_khinchine_wait, service_scv_from_tokens
+from .mg1 import MG1Corrector, MG1Metrics, auto_scv_enabled, configured_scv, pollaczek_khinchine_wait, recommended_service_scv, service_scv_from_tokens
 from .search import BelowRegion, InRegion, AboveRegion, binary_search, within_tolerance
 from .queueanalyzer import (
     EPSILON,
@@ -54,6 +54,8 @@ __all__ = [
     "effective_concurrency",
     "MG1Corrector",
     "configured_scv",
+    "auto_scv_enabled",
+    "recommended_service_scv",
     "MG1Metrics",
     "pollaczek_khinchine_wait",
     "service_scv_from_tokens",

@@ -46,6 +46,8 @@ __all__ = [
     "MG1Corrector",
     "MG1Metrics",
     "configured_scv",
+    "auto_scv_enabled",
+    "recommended_service_scv",
 ]
 
 
@@ -63,12 +65,48 @@ def configured_scv() -> float:
     if mode != "mg1":
         raise ValueError(f"unknown WVA_ANALYZER mode {mode!r} (use 'mm1k' or 'mg1')")
     raw = os.environ.get("WVA_SERVICE_SCV", "").strip()
-    if not raw:
+    if not raw or raw.lower() == "auto":
+        # "auto": per-server values are estimated from the serving
+        # engine's token histogram (collector.estimate_token_scv) and
+        # override this fleet-wide fallback
         return 0.5
     scv = float(raw)
     if scv < 0:
         raise ValueError(f"WVA_SERVICE_SCV must be >= 0, got {scv}")
     return scv
+
+
+def auto_scv_enabled() -> bool:
+    """True when mg1 mode should derive cs^2 per model from the measured
+    token histogram (WVA_ANALYZER=mg1 + WVA_SERVICE_SCV=auto)."""
+    return (
+        os.environ.get("WVA_ANALYZER", "").strip().lower() == "mg1"
+        and os.environ.get("WVA_SERVICE_SCV", "").strip().lower() == "auto"
+    )
+
+
+def recommended_service_scv(
+    mean_out_tokens: float,
+    token_scv: float,
+    alpha: float,
+    beta: float,
+    gamma: float,
+    delta: float,
+    in_tokens: float,
+) -> float:
+    """Service-time cs^2 from measured token-count variability under the
+    batch-1 service law S = gamma + delta*in + (K-1)(alpha + beta): the
+    affine offset damps the token SCV (service_scv_from_tokens).  An
+    approximation (batch-dependent terms shift mean and deviation the
+    same way to first order), clamped to [0, 1] so auto mode never sizes
+    less conservatively than exponential."""
+    scv = service_scv_from_tokens(
+        mean_out_tokens=max(mean_out_tokens - 1.0, 0.0),
+        scv_out_tokens=max(token_scv, 0.0),
+        decode_time=alpha + beta,
+        fixed_time=gamma + delta * max(in_tokens, 0.0),
+    )
+    return min(max(scv, 0.0), 1.0)
 
 
 def pollaczek_khinchine_wait(arrival_rate: float, service_time: float, scv: float) -> float:

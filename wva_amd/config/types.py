@@ -209,6 +209,12 @@ class ServerSpec(SpecBase):
     max_batch_size: int = jfield("maxBatchSize", 0)  # override
     current_alloc: AllocationData = jfield("currentAlloc", default_factory=AllocationData)
     desired_alloc: AllocationData = jfield("desiredAlloc", default_factory=AllocationData)
+    # MI355X extension (not in the reference spec JSON): per-server
+    # service-time cs^2 for the mg1 analyzer; negative = unset -> the
+    # global WVA_SERVICE_SCV applies (0 is a VALID measured value:
+    # deterministic lengths).  Populated from the measured token
+    # histogram in auto mode (collector.estimate_token_scv).
+    service_scv: float = jfield("serviceSCV", -1.0)
 
 
 @dataclass

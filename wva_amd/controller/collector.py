@@ -276,3 +276,69 @@ def collect_inventory_k8s(client) -> Dict[str, Dict[str, object]]:
             entry["count"] += count
             entry["nodes"] += 1
     return out
+
+
+# --------------------------------------------------- token-variance estimate
+def token_scv_query(model: str, namespace: str) -> str:
+    """Per-le bucket rates of the generation-token histogram."""
+    return (
+        f'rate(vllm:request_generation_tokens_bucket'
+        f'{{{constants.LABEL_MODEL_NAME}="{model}",'
+        f'{constants.LABEL_NAMESPACE}="{namespace}"}}[{rate_window()}])'
+    )
+
+
+def estimate_token_scv(prom: PromAPI, model: str, namespace: str):
+    """Squared coefficient of variation of output-token counts, estimated
+    from the ``vllm:request_generation_tokens`` histogram the serving
+    engine already exports (the reference's emulator omits this series;
+    ours emits it).
+
+    Cumulative bucket rates -> per-bucket probabilities -> first/second
+    moments with bucket midpoints (+Inf capped at 1.5x the last finite
+    edge).  Returns None when the histogram is absent or the window is
+    empty — callers fall back to the configured cs^2.  This is what
+    makes ``WVA_ANALYZER=mg1`` + ``WVA_SERVICE_SCV=auto`` self-
+    configuring: measured length variability, not an operator guess.
+    """
+    try:
+        vec = prom.query(token_scv_query(model, namespace))
+    except PromQueryError:
+        return None
+    if not vec:
+        # emulator fallback: no namespace label
+        try:
+            vec = prom.query(
+                f'rate(vllm:request_generation_tokens_bucket'
+                f'{{{constants.LABEL_MODEL_NAME}="{model}"}}[{rate_window()}])'
+            )
+        except PromQueryError:
+            return None
+    buckets = []
+    for s in vec:
+        le = s.labels.get("le", "")
+        if not le:
+            continue
+        edge = math.inf if le in ("+Inf", "Inf", "inf") else float(le)
+        buckets.append((edge, max(_fix_value(s.value), 0.0)))
+    if len(buckets) < 2:
+        return None
+    buckets.sort(key=lambda b: b[0])
+    total = buckets[-1][1]
+    if total <= 0:
+        return None
+    last_finite = max((e for e, _ in buckets if math.isfinite(e)), default=0.0)
+    m1 = 0.0
+    m2 = 0.0
+    prev_edge, prev_cum = 0.0, 0.0
+    for edge, cum in buckets:
+        p = max(cum - prev_cum, 0.0) / total
+        hi = edge if math.isfinite(edge) else last_finite * 1.5
+        mid = 0.5 * (prev_edge + hi)
+        m1 += p * mid
+        m2 += p * mid * mid
+        prev_edge, prev_cum = (hi if math.isfinite(edge) else prev_edge), cum
+    if m1 <= 0:
+        return None
+    var = max(m2 - m1 * m1, 0.0)
+    return var / (m1 * m1)

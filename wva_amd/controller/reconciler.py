@@ -420,6 +420,42 @@ class VariantAutoscalingReconciler:
                 log.info("variantAutoscaling bad deployment server data, skipping", name=update_va.name)
                 continue
 
+            # mg1 auto mode: per-model cs^2 from the measured generation-
+            # token histogram (never fatal; absent histogram -> the
+            # fleet-wide WVA_SERVICE_SCV fallback applies)
+            from ..analyzer.mg1 import auto_scv_enabled, recommended_service_scv
+
+            if auto_scv_enabled():
+                try:
+                    token_scv = collector.estimate_token_scv(
+                        self.prom_api, model_name, update_va.namespace
+                    )
+                    if token_scv is not None:
+                        profile = update_va.spec.model_profile.accelerators[0]
+                        pp = profile.perf_parms
+                        scv = recommended_service_scv(
+                            mean_out_tokens=float(
+                                update_va.status.current_alloc.load.avg_output_tokens or 0
+                            ),
+                            token_scv=token_scv,
+                            alpha=float(pp.decode_parms.get("alpha", "0")),
+                            beta=float(pp.decode_parms.get("beta", "0")),
+                            gamma=float(pp.prefill_parms.get("gamma", "0")),
+                            delta=float(pp.prefill_parms.get("delta", "0")),
+                            in_tokens=float(
+                                update_va.status.current_alloc.load.avg_input_tokens or 0
+                            ),
+                        )
+                        system_data.spec.servers.spec[-1].service_scv = scv
+                        log.info(
+                            "auto service cs^2 from token histogram",
+                            variant=update_va.name,
+                            token_scv=round(token_scv, 4),
+                            service_scv=round(scv, 4),
+                        )
+                except Exception as e:
+                    log.warn("token-scv estimation failed; using fallback", error=str(e))
+
             update_list.append(update_va)
             va_map[full_name(va.name, va.namespace)] = va
         return update_list, va_map, responses

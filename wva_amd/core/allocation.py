@@ -198,7 +198,14 @@ def create_allocation(system: "System", server_name: str, acc_name: str) -> Opti
     try:
         from ..analyzer.mg1 import configured_scv
 
-        qa = QueueAnalyzer(config, request_size, scv=configured_scv())
+        # per-server cs^2 (mg1 auto mode, measured token histogram)
+        # overrides the fleet-wide setting; negative = unset
+        server_scv = getattr(server, "service_scv", -1.0)
+        qa = QueueAnalyzer(
+            config,
+            request_size,
+            scv=server_scv if server_scv >= 0 else configured_scv(),
+        )
     except AnalyzerError:
         return None
 

@@ -30,6 +30,8 @@ class Server:
         self.keep_accelerator = spec.keep_accelerator
         self.min_num_replicas = spec.min_num_replicas
         self.max_batch_size = spec.max_batch_size
+        # negative = unset -> the global analyzer configuration applies
+        self.service_scv = getattr(spec, "service_scv", -1.0)
         self.load: Optional[ServerLoadSpec] = spec.current_alloc.load
         self.all_allocations: Dict[str, Allocation] = {}
         self.allocation: Optional[Allocation] = None

@@ -109,8 +109,12 @@ def _graph_solve(native, problems_t, max_k):
             return None
 
 
-def _solve_problems_python(problems: np.ndarray, scv: float = 1.0) -> np.ndarray:
-    """Reference-semantics scalar fallback via the Python analyzer."""
+def _solve_problems_python(problems: np.ndarray, scv=1.0) -> np.ndarray:
+    """Reference-semantics scalar fallback via the Python analyzer.
+
+    ``scv`` is the service-time cs^2 — a scalar for the whole batch or a
+    per-row array (mg1 auto mode derives one per server)."""
+    scv_arr = np.broadcast_to(np.asarray(scv, dtype=np.float64), (problems.shape[0],))
     out = np.zeros((problems.shape[0], RESULT_FIELDS), dtype=np.float64)
     for i, row in enumerate(problems):
         N = int(row[P_MAX_BATCH])
@@ -129,7 +133,7 @@ def _solve_problems_python(problems: np.ndarray, scv: float = 1.0) -> np.ndarray
                     avg_input_tokens=int(row[P_IN_TOKENS]),
                     avg_output_tokens=int(row[P_OUT_TOKENS]),
                 ),
-                scv=scv,
+                scv=float(scv_arr[i]),
             )
             _, metrics, _ = qa.size(
                 TargetPerf(
@@ -273,7 +277,9 @@ class BatchedAllocationSolver:
         keys: List[Tuple[str, str, int]] = []  # (server, acc, N)
         zero_load: Dict[str, Dict[str, Allocation]] = {}
         costs: List[float] = []
+        scv_rows: List[float] = []  # per-server cs^2 (negative = use global)
         rows_append, keys_append, costs_append = rows.append, keys.append, costs.append
+        scv_append = scv_rows.append
 
         for server in system.servers.values():
             server.all_allocations = {}
@@ -321,9 +327,22 @@ class BatchedAllocationSolver:
                 )
                 keys_append((server_name, acc.name, N))
                 costs_append(cost_per_rep)
+                scv_append(getattr(server, "service_scv", -1.0))
 
         if rows:
-            results = solve_problems(np.array(rows, dtype=np.float64), self.device)
+            arr = np.array(rows, dtype=np.float64)
+            from ..analyzer.mg1 import configured_scv
+
+            global_scv = configured_scv()
+            scv_arr = np.array(
+                [s if s >= 0 else global_scv for s in scv_rows], dtype=np.float64
+            )
+            if (scv_arr != 1.0).any():
+                # M/G/1-corrected sizing (global mg1 mode and/or
+                # per-server measured cs^2): the scalar analyzer path
+                results = _solve_problems_python(arr, scv=scv_arr)
+            else:
+                results = solve_problems(arr, self.device)
         else:
             results = np.zeros((0, RESULT_FIELDS))
 
