@@ -202,3 +202,100 @@ class TestBinarySearchPrecision:
         assert ind == InRegion
         _, ind = binary_search(1.0, 5.0, 10.0, fn)
         assert ind == InRegion
+
+
+def _valid_parms():
+    return ServiceParms(
+        prefill=PrefillParms(gamma=10.0, delta=0.001),
+        decode=DecodeParms(alpha=1.0, beta=0.01),
+    )
+
+
+class TestConfigurationCheckTable:
+    # queueanalyzer_test.go:92 TestConfiguration_Check
+    @pytest.mark.parametrize(
+        "name,max_batch,max_queue,parms,want_err",
+        [
+            ("valid configuration", 8, 16, "valid", False),
+            ("zero max batch size", 0, 16, "valid", True),
+            ("negative max batch size", -1, 16, "valid", True),
+            ("negative max queue size", 8, -1, "valid", True),
+            ("nil service parameters", 8, 16, None, True),
+        ],
+    )
+    def test_table(self, name, max_batch, max_queue, parms, want_err):
+        from wva_amd.analyzer import AnalyzerError
+
+        config = Configuration(
+            max_batch_size=max_batch,
+            max_queue_size=max_queue,
+            service_parms=_valid_parms() if parms == "valid" else None,
+        )
+        rs = RequestSize(avg_input_tokens=100, avg_output_tokens=10)
+        if want_err:
+            with pytest.raises(AnalyzerError):
+                QueueAnalyzer(config, rs)
+        else:
+            assert QueueAnalyzer(config, rs) is not None
+
+
+class TestRequestSizeCheckTable:
+    # queueanalyzer_test.go:178 TestRequestSize_Check
+    @pytest.mark.parametrize(
+        "in_tok,out_tok,want_err",
+        [
+            (100, 10, False),
+            (0, 10, False),      # zero input = decode-only, valid
+            (-1, 10, True),
+            (100, 0, True),      # must produce at least one token
+            (100, -5, True),
+        ],
+    )
+    def test_table(self, in_tok, out_tok, want_err):
+        from wva_amd.analyzer import AnalyzerError
+
+        config = Configuration(
+            max_batch_size=8, max_queue_size=16, service_parms=_valid_parms()
+        )
+        if want_err:
+            with pytest.raises(AnalyzerError):
+                QueueAnalyzer(
+                    config,
+                    RequestSize(avg_input_tokens=in_tok, avg_output_tokens=out_tok),
+                )
+        else:
+            QueueAnalyzer(
+                config, RequestSize(avg_input_tokens=in_tok, avg_output_tokens=out_tok)
+            )
+
+
+class TestPrefillTimeTable:
+    # queueanalyzer_test.go:226 TestPrefillParms_PrefillTime
+    @pytest.mark.parametrize(
+        "name,in_tok,batch,expected",
+        [
+            ("no input tokens", 0, 4.0, 0.0),
+            ("small batch", 1000, 1.0, 11.0),
+            ("large batch", 2000, 8.0, 26.0),
+            ("fractional batch size", 500, 2.5, 11.25),
+        ],
+    )
+    def test_table(self, name, in_tok, batch, expected):
+        prefill = PrefillParms(gamma=10.0, delta=0.001)
+        assert prefill.prefill_time(in_tok, batch) == pytest.approx(expected)
+
+
+class TestDecodeTimeTable:
+    # queueanalyzer_test.go:274 TestDecodeParms_DecodeTime
+    @pytest.mark.parametrize(
+        "batch,expected",
+        [
+            (1.0, 1.01),
+            (8.0, 1.08),
+            (0.0, 1.0),
+            (2.5, 1.025),
+        ],
+    )
+    def test_table(self, batch, expected):
+        decode = DecodeParms(alpha=1.0, beta=0.01)
+        assert decode.decode_time(batch) == pytest.approx(expected)
