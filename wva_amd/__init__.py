@@ -25,4 +25,4 @@ samples and the bundled vLLM emulator are CDNA4-first (MI355X: 288 GB HBM3E,
 ~8 TB/s, 1.4 kW OAM envelope).
 """
 
-__version__ = "0.1.0"
+__version__ = "0.2.0"  # round 2
