@@ -381,3 +381,34 @@ class TestAutoScvEstimation:
             f"replicas than mm1k on a wait-bound workload "
             f"(mm1k={markovian}, auto={auto})"
         )
+
+
+class TestPollaczekKhinchineCrossValidation:
+    @pytest.mark.parametrize("scv", [0.0, 0.5, 1.0])
+    @pytest.mark.parametrize("rho", [0.3, 0.6, 0.8])
+    def test_corrected_analyzer_matches_pk_formula(self, scv, rho):
+        """At batch size 1 with a deep queue, the state-dependent chain
+        IS an M/M/1(/K) queue; the Allen-Cunneen-corrected wait must
+        match the exact Pollaczek-Khinchine M/G/1 wait."""
+        from wva_amd.analyzer import pollaczek_khinchine_wait
+
+        alpha, beta = 10.0, 0.0   # constant service
+        out_tokens = 11           # 10 decode passes
+        service_ms = (out_tokens - 1) * alpha  # 100 ms
+        config = Configuration(
+            max_batch_size=1,
+            max_queue_size=2000,  # deep enough to approximate infinite K
+            service_parms=ServiceParms(
+                prefill=PrefillParms(gamma=0.0, delta=0.0),
+                decode=DecodeParms(alpha=alpha, beta=beta),
+            ),
+        )
+        qa = QueueAnalyzer(
+            config,
+            RequestSize(avg_input_tokens=0, avg_output_tokens=out_tokens),
+            scv=scv,
+        )
+        lam_per_ms = rho / service_ms
+        got = qa.analyze(lam_per_ms * 1000.0).avg_wait_time
+        want = pollaczek_khinchine_wait(lam_per_ms, service_ms, scv)
+        assert got == pytest.approx(want, rel=2e-3), (scv, rho)
