@@ -522,3 +522,72 @@ class TestSeenUidCompaction:
         finally:
             stop.set()
             t.join(timeout=8)
+
+
+class TestEventLogStateMachine:
+    """Stateful fuzz of the watch-event log (the stub's source of truth):
+    resourceVersions strictly increase, events_since(rv) returns exactly
+    the suffix after rv, and GoneError fires iff rv predates the
+    retained window."""
+
+    def test_random_operation_sequences(self):
+        import random
+
+        from wva_amd.api.v1alpha1.types import ObjectMeta
+        from wva_amd.kube import Deployment, InMemoryKubeClient
+        from wva_amd.kube.errors import GoneError, NotFoundError
+
+        rng = random.Random(421)
+        store = InMemoryKubeClient(event_log_limit=16)
+        shadow = []  # (rv, type, name) mirror of every event ever
+        live = set()
+
+        def record_all():
+            # ground truth from an rv=0 replay is impossible once
+            # compacted; track via shadow appended on each op
+            pass
+
+        for step in range(400):
+            op = rng.random()
+            name = f"obj-{rng.randrange(12)}"
+            try:
+                if op < 0.5:
+                    obj = Deployment(metadata=ObjectMeta(name=name, namespace="ns"))
+                    created = store.create(obj)
+                    shadow.append((created.metadata.resource_version, "ADDED", name))
+                    live.add(name)
+                elif op < 0.8:
+                    got = store.get(Deployment, name, "ns")
+                    updated = store.update(got)
+                    shadow.append((updated.metadata.resource_version, "MODIFIED", name))
+                else:
+                    store.delete(Deployment, name, "ns")
+                    live.discard(name)
+                    shadow.append((store.resource_version, "DELETED", name))
+            except (NotFoundError, Exception) as e:
+                if type(e).__name__ not in ("NotFoundError", "ConflictError"):
+                    raise
+
+            # invariant: shadow rvs strictly increase
+            rvs = [rv for rv, _, _ in shadow]
+            assert rvs == sorted(rvs) and len(set(rvs)) == len(rvs)
+
+            # invariant: events_since agrees with the shadow suffix for
+            # any rv inside the retained window; GoneError outside it
+            if shadow:
+                probe_rv = rng.choice(
+                    [0, shadow[0][0] - 1, shadow[len(shadow) // 2][0], shadow[-1][0]]
+                )
+                try:
+                    events = store.events_since(probe_rv)
+                    got = [(rv, et, o.metadata.name) for rv, et, o in events]
+                    want = [e for e in shadow if e[0] > probe_rv]
+                    # only comparable when the suffix is fully retained
+                    if probe_rv >= store._compacted_to:
+                        assert got == want, (probe_rv, got[:3], want[:3])
+                except GoneError:
+                    assert probe_rv < store._compacted_to
+
+            # invariant: list matches the live set
+            names = {d.metadata.name for d in store.list(Deployment, "ns")}
+            assert names == live
