@@ -587,3 +587,49 @@ class TestGreedyCapacityInvariants:
         if high.allocation is None:
             # feasible-for-neither is the only excuse
             assert low.allocation is None
+
+
+class TestRoundRobinEqualSplit:
+    def test_two_equal_servers_split_the_pool(self):
+        """Deterministic round-robin outcome: two identical servers each
+        wanting 4 replicas on a 4-unit pool end with 2 each (ticket
+        cycling grants one replica per turn)."""
+        system, _ = greedy_system()
+        available = {MI300X_T: 4, MI355X_T: 0}
+        s1, s3 = system.servers["server1"], system.servers["server3"]
+        s1.remove_allocation()
+        s3.remove_allocation()
+        # both llama-8b servers restricted to their MI300X candidate,
+        # each wanting 4 replicas on a 4-unit pool
+        e1 = ServerEntry("server1", 1, [
+            a for n, a in s1.all_allocations.items() if n == "MI300X"
+        ])
+        e3 = ServerEntry("server3", 1, [
+            a for n, a in s3.all_allocations.items() if n == "MI300X"
+        ])
+        assert e1.allocations and e3.allocations
+        e1.allocations[0].num_replicas = 4
+        e3.allocations[0].num_replicas = 4
+        _allocate_equally(system, [e1, e3], available)
+        a1, a3 = s1.allocation, s3.allocation
+        assert a1 is not None and a3 is not None
+        # identical unit weights: the round-robin tickets split evenly
+        assert a1.num_replicas == 2 and a3.num_replicas == 2
+        assert available[MI300X_T] == 0
+
+    def test_single_server_gets_partial_grant(self):
+        # wants 4 replicas, pool has 2 units -> best effort grants 2
+        system, _ = greedy_system()
+        available = {MI300X_T: 2, MI355X_T: 0}
+        s1 = system.servers["server1"]
+        s1.remove_allocation()
+        e1 = ServerEntry("server1", 1, [
+            a for n, a in s1.all_allocations.items() if n == "MI300X"
+        ])
+        e1.allocations[0].num_replicas = 4
+        want_cost = e1.allocations[0].cost
+        _allocate_equally(system, [e1], available)
+        a1 = s1.allocation
+        assert a1 is not None
+        assert a1.num_replicas == 2
+        assert available[MI300X_T] == 0
