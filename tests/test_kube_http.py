@@ -204,3 +204,56 @@ class TestWatch:
             runtime.stop()
         finally:
             ctrl_metrics.reset_metrics()
+
+
+class TestOptimizationFailureOverConformantStub:
+    def test_first_cycle_failure_tolerates_status_rejection(self):
+        """First-cycle optimization failure writes OptimizationReady=False
+        to VAs whose desiredOptimizedAlloc is still empty; the conformant
+        stub rejects that write (as a real apiserver would: minLength on
+        the accelerator) and the reconciler must log-and-continue —
+        reference semantics (variantautoscaling_controller.go:168-186,
+        the statusErr is only logged).  Fresh stub: the failure must be
+        fleet-wide (engine errors only when NO variant is feasible)."""
+        import uvicorn
+
+        from wva_amd.controller import metrics as ctrl_metrics
+        from kube_fixtures import make_cluster, make_deployment, make_va, set_load_metrics
+        from wva_amd.kube.http_client import HTTPKubeClient
+
+        store = make_cluster()
+        app, _ = create_stub_api_server(store)
+        server = uvicorn.Server(
+            uvicorn.Config(app, host="127.0.0.1", port=0, log_level="error")
+        )
+        t = threading.Thread(target=server.run, daemon=True)
+        t.start()
+        for _ in range(200):
+            if server.started:
+                break
+            time.sleep(0.05)
+        port = server.servers[0].sockets[0].getsockname()[1]
+        client = HTTPKubeClient(base_url=f"http://127.0.0.1:{port}", token="t")
+        make_deployment(store, name="fail-va", replicas=1)
+        # impossible ITL target: alpha alone exceeds it -> no feasible
+        # allocation anywhere -> engine error
+        make_va(store, name="fail-va", model_id="default/llama-8b",
+                max_batch=8, alpha="500.0", beta="1.0")
+        prom = MockPromAPI()
+        set_load_metrics(prom, "default/llama-8b", "default",
+                         arrival_rps=5.0, out_tokens=100.0)
+        registry = CollectorRegistry()
+        ctrl_metrics.init_metrics(registry)
+        try:
+            rec = VariantAutoscalingReconciler(client, prom)
+            result = rec.reconcile()  # must not raise
+            assert result.requeue_after is not None
+        finally:
+            ctrl_metrics.reset_metrics()
+        va = store.get(v1alpha1.VariantAutoscaling, "fail-va", "default")
+        # the write was rejected wholesale: desired stays empty and the
+        # condition did not land (matching real-apiserver behavior)
+        assert va.status.desired_optimized_alloc.num_replicas == 0
+        assert not v1alpha1.is_condition_true(va, v1alpha1.TYPE_OPTIMIZATION_READY)
+        server.should_exit = True
+        t.join(timeout=5)
