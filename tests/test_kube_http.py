@@ -257,3 +257,48 @@ class TestOptimizationFailureOverConformantStub:
         assert not v1alpha1.is_condition_true(va, v1alpha1.TYPE_OPTIMIZATION_READY)
         server.should_exit = True
         t.join(timeout=5)
+
+
+class TestOwnerRefGCOverHTTP:
+    def test_va_garbage_collected_when_deployment_deleted_via_api(self):
+        """Single-VA lifecycle over the wire (e2e_test.go:630): the
+        reconciler sets the ownerReference through a PATCH, and deleting
+        the owning Deployment via the API cascades to the VA."""
+        import uvicorn
+
+        from kube_fixtures import make_cluster, make_deployment, make_va, set_load_metrics
+
+        store = make_cluster()
+        app, _ = create_stub_api_server(store)
+        server = uvicorn.Server(
+            uvicorn.Config(app, host="127.0.0.1", port=0, log_level="error")
+        )
+        t = threading.Thread(target=server.run, daemon=True)
+        t.start()
+        for _ in range(200):
+            if server.started:
+                break
+            time.sleep(0.05)
+        port = server.servers[0].sockets[0].getsockname()[1]
+        client = HTTPKubeClient(base_url=f"http://127.0.0.1:{port}")
+        try:
+            make_deployment(store, name="gc-va", replicas=1)
+            make_va(store, name="gc-va")
+            prom = MockPromAPI()
+            set_load_metrics(prom, "default/llama-8b", "default", arrival_rps=2.0)
+            registry = CollectorRegistry()
+            ctrl_metrics.init_metrics(registry)
+            try:
+                VariantAutoscalingReconciler(client, prom).reconcile()
+            finally:
+                ctrl_metrics.reset_metrics()
+            va = client.get(v1alpha1.VariantAutoscaling, "gc-va", "default")
+            assert any(r.kind == "Deployment" and r.name == "gc-va"
+                       for r in va.metadata.owner_references)
+            # delete the owner THROUGH the API; GC cascades to the VA
+            client.delete(Deployment, "gc-va", "default")
+            with pytest.raises(NotFoundError):
+                client.get(v1alpha1.VariantAutoscaling, "gc-va", "default")
+        finally:
+            server.should_exit = True
+            t.join(timeout=5)
