@@ -484,3 +484,23 @@ class TestMainInClusterStartup:
             kube.should_exit = True
             prom_t.join(timeout=5)
             kube_t.join(timeout=5)
+
+
+class TestLimitedModeOverlay:
+    def test_overlay_parses_and_patches_the_right_configmap(self):
+        import yaml as _yaml
+
+        base = Path(__file__).resolve().parent.parent / "deploy" / "limited-mode"
+        kust = _yaml.safe_load((base / "kustomization.yaml").read_text())
+        assert kust["resources"] == ["../"]
+        patch = _yaml.safe_load((base / "configmap-patch.yaml").read_text())
+        assert patch["metadata"]["name"] == CONFIG_MAP_NAME
+        assert patch["metadata"]["namespace"] == CONFIG_MAP_NAMESPACE
+        assert patch["data"]["WVA_OPTIMIZER_MODE"] == "limited"
+        assert patch["data"]["WVA_INVENTORY"] == "k8s"
+        # the patched keys are ones the controller actually reads
+        from wva_amd.controller.utils import create_system_data
+
+        sd = create_system_data({}, {}, patch["data"])
+        assert sd.spec.optimizer.spec.unlimited is False
+        assert sd.spec.optimizer.spec.saturation_policy == "PriorityRoundRobin"
