@@ -226,3 +226,34 @@ class TestGreedyPriorityProperties:
         more = replicas(h)
         if base is not None and more is not None:
             assert more >= base
+
+
+class TestMG1Properties:
+    @settings(max_examples=20, deadline=None)
+    @given(parms, st.floats(0.05, 0.95), st.floats(0.0, 1.0), st.floats(0.0, 1.0))
+    def test_wait_monotone_in_scv(self, p, frac, s_lo, s_hi):
+        """Corrected waiting time is monotone in cs^2 at any operating
+        point, and the cs^2=1 analyzer is bitwise the Markovian one."""
+        from wva_amd.analyzer import Configuration, QueueAnalyzer, RequestSize
+
+        lo, hi = sorted((s_lo, s_hi))
+        config = Configuration(
+            max_batch_size=p["max_batch"],
+            max_queue_size=p["max_batch"] * 10,
+            service_parms=ServiceParms(
+                prefill=PrefillParms(p["gamma"], p["delta"]),
+                decode=DecodeParms(p["alpha"], p["beta"]),
+            ),
+        )
+        rs = RequestSize(p["in_tokens"], p["out_tokens"])
+        qa_lo = QueueAnalyzer(config, rs, scv=lo)
+        qa_hi = QueueAnalyzer(config, rs, scv=hi)
+        qa_1 = QueueAnalyzer(config, rs, scv=1.0)
+        qa_ref = QueueAnalyzer(config, rs)
+        rate = qa_ref.rate_range.min + frac * (
+            qa_ref.rate_range.max - qa_ref.rate_range.min
+        )
+        w_lo = qa_lo.analyze(rate).avg_wait_time
+        w_hi = qa_hi.analyze(rate).avg_wait_time
+        assert w_lo <= w_hi + 1e-12
+        assert qa_1.analyze(rate).avg_wait_time == qa_ref.analyze(rate).avg_wait_time
