@@ -4,8 +4,10 @@ The reference sizes one (server, accelerator) pair at a time inside the
 reconcile loop (hot loop #1: /root/reference/pkg/core/server.go:55-67 ->
 allocation.go:27-163).  Here the analyze phase is *batched*: all pairs of
 the fleet are packed into one [B, 12] float64 matrix and solved either by
-the native extension (CPU C++/at::parallel_for, or the gfx950 HIP kernel —
-one workgroup per pair) or by the pure-Python analyzer fallback.
+the native extension (the torch-free C++/OpenMP binding on CPU, or the
+gfx950 HIP kernel — one workgroup per pair) or by the pure-Python
+analyzer fallback; mg1 mode routes through the scalar analyzer with
+per-server cs^2 support.
 
 Field layouts mirror wva_amd/csrc/queue_core.h.
 """
