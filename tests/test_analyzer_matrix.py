@@ -299,3 +299,48 @@ class TestDecodeTimeTable:
     def test_table(self, batch, expected):
         decode = DecodeParms(alpha=1.0, beta=0.01)
         assert decode.decode_time(batch) == pytest.approx(expected)
+
+
+class TestMM1KUtilizationGrid:
+    # queuemodel_test.go:152 TestMM1KModel_ProbabilityCalculation — the
+    # utilization grid incl. the lambda == mu boundary
+    @pytest.mark.parametrize(
+        "name,lam,mu",
+        [
+            ("low utilization", 0.5, 2.0),
+            ("medium utilization", 1.5, 2.0),
+            ("high utilization", 1.9, 2.0),
+            ("equal arrival and service rates", 2.0, 2.0),
+        ],
+    )
+    def test_probability_grid(self, name, lam, mu):
+        from wva_amd.analyzer import MM1KModel
+
+        model = MM1KModel(3)
+        model.solve(lam, mu)
+        if not model.is_valid:
+            pytest.skip("invalid model for this point")
+        assert (model.p >= 0).all()
+        assert float(model.p.sum()) == pytest.approx(1.0, abs=1e-9)
+        assert 0.0 <= model.throughput <= lam
+
+
+class TestMM1KvsStateDependentComparison:
+    # queuemodel_test.go:461 TestMM1Models_Comparison: constant service
+    # rates must make the state-dependent chain coincide with M/M/1/K
+    def test_constant_rates_coincide(self):
+        import numpy as np
+
+        from wva_amd.analyzer import MM1KModel, MM1ModelStateDependent
+
+        K, rate, lam = 5, 3.0, 1.5
+        mm1k = MM1KModel(K)
+        mm1k.solve(lam, rate)
+        statedep = MM1ModelStateDependent(K, np.full(K, rate))
+        statedep.solve(lam, 1.0)
+        assert mm1k.is_valid and statedep.is_valid
+        assert statedep.avg_num_in_system == pytest.approx(
+            mm1k.avg_num_in_system, abs=1e-9
+        )
+        assert statedep.throughput == pytest.approx(mm1k.throughput, abs=1e-9)
+        assert statedep.avg_resp_time == pytest.approx(mm1k.avg_resp_time, abs=1e-9)
