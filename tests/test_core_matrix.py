@@ -248,3 +248,87 @@ class TestServerCandidateMatrix:
         server = system.server("all")
         cands = server.get_candidate_accelerators(system.accelerators)
         assert set(cands) == set(system.accelerators)
+
+
+class TestModelPerfDataTable:
+    """model_test.go:9-120 — Model add/remove/get perf-data table."""
+
+    def _spec(self, name="m", acc="MI355X", count=1):
+        return ModelAcceleratorPerfData(
+            name=name, acc=acc, acc_count=count, max_batch_size=8,
+            decode_parms=DecodeParmsSpec(alpha="10", beta="5"),
+            prefill_parms=PrefillParmsSpec(gamma="20", delta="0.1"),
+        )
+
+    def test_valid_perf_data(self):
+        from wva_amd.core.model import Model
+
+        m = Model("m")
+        m.add_perf_data(self._spec(count=2))
+        assert m.get_perf_data("MI355X") is not None
+        assert m.get_num_instances("MI355X") == 2
+
+    def test_zero_accelerator_count_defaults_to_one(self):
+        from wva_amd.core.model import Model
+
+        m = Model("m")
+        m.add_perf_data(self._spec(count=0))
+        assert m.get_num_instances("MI355X") == 1
+
+    def test_negative_accelerator_count_defaults_to_one(self):
+        from wva_amd.core.model import Model
+
+        m = Model("m")
+        m.add_perf_data(self._spec(count=-3))
+        assert m.get_num_instances("MI355X") == 1
+
+    def test_wrong_model_spec_ignored(self):
+        # model_test.go:108 — a spec for a different model is not absorbed
+        from wva_amd.core.model import Model
+
+        m = Model("m")
+        m.add_perf_data(self._spec(name="other"))
+        assert m.get_perf_data("MI355X") is None
+        assert m.get_num_instances("MI355X") == 0
+
+    def test_remove_perf_data(self):
+        from wva_amd.core.model import Model
+
+        m = Model("m")
+        m.add_perf_data(self._spec())
+        m.remove_perf_data("MI355X")
+        assert m.get_perf_data("MI355X") is None
+        # removing a missing accelerator is a no-op, not an error
+        m.remove_perf_data("MI300X")
+
+
+class TestSaturationPolicyEnum:
+    """config_test.go:7-110 — policy string/parse round-trip table."""
+
+    @pytest.mark.parametrize(
+        "policy_str,expected",
+        [
+            ("None", "NONE"),
+            ("PriorityExhaustive", "PRIORITY_EXHAUSTIVE"),
+            ("PriorityRoundRobin", "PRIORITY_ROUND_ROBIN"),
+            ("RoundRobin", "ROUND_ROBIN"),
+        ],
+    )
+    def test_parse_known(self, policy_str, expected):
+        from wva_amd.config.policies import SaturationPolicy
+
+        assert SaturationPolicy.parse(policy_str).name == expected
+
+    @pytest.mark.parametrize("bad", ["Unknown", "", "priorityexhaustive", "NONE"])
+    def test_parse_unknown_returns_default(self, bad):
+        # config_test.go:76-96 — unknown / empty / wrong-case → default
+        from wva_amd.config.policies import DEFAULT_SATURATION_POLICY, SaturationPolicy
+
+        assert SaturationPolicy.parse(bad) is DEFAULT_SATURATION_POLICY
+
+    def test_round_trip(self):
+        # config_test.go:101 — String() ∘ parse() is identity for all members
+        from wva_amd.config.policies import SaturationPolicy
+
+        for p in SaturationPolicy:
+            assert SaturationPolicy.parse(str(p)) is p
