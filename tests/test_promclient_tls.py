@@ -184,3 +184,32 @@ class TestTLSValidation:
         assert config.base_url == "https://env:9090"
         assert config.insecure_skip_verify is True
         assert config.bearer_token == "tok"
+
+    def test_env_parsing_openshift_matrix(self, monkeypatch):
+        # tls_test.go:86-117 — the full OpenShift thanos-querier env shape:
+        # every field lands, empty client cert/key stay empty
+        from wva_amd.controller.promclient import parse_prometheus_config_from_env
+
+        monkeypatch.setenv(
+            "PROMETHEUS_BASE_URL",
+            "https://thanos-querier.openshift-monitoring.svc.cluster.local:9091",
+        )
+        monkeypatch.setenv("PROMETHEUS_TLS_INSECURE_SKIP_VERIFY", "false")
+        monkeypatch.setenv("PROMETHEUS_CA_CERT_PATH", "/etc/openshift-ca/ca.crt")
+        monkeypatch.setenv("PROMETHEUS_CLIENT_CERT_PATH", "")
+        monkeypatch.setenv("PROMETHEUS_CLIENT_KEY_PATH", "")
+        monkeypatch.setenv(
+            "PROMETHEUS_SERVER_NAME", "thanos-querier.openshift-monitoring.svc"
+        )
+        monkeypatch.setenv(
+            "PROMETHEUS_TOKEN_PATH",
+            "/var/run/secrets/kubernetes.io/serviceaccount/token",
+        )
+        config = parse_prometheus_config_from_env()
+        assert config.base_url.endswith(":9091")
+        assert config.insecure_skip_verify is False
+        assert config.ca_cert_path == "/etc/openshift-ca/ca.crt"
+        assert config.client_cert_path == ""
+        assert config.client_key_path == ""
+        assert config.server_name == "thanos-querier.openshift-monitoring.svc"
+        assert config.token_path == "/var/run/secrets/kubernetes.io/serviceaccount/token"
