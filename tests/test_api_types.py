@@ -181,3 +181,21 @@ class TestDeepCopyIndependence:
         d = va.model_dump(by_alias=True, exclude_none=True, mode="json")
         raw = d["status"]["desiredOptimizedAlloc"]["lastRunTime"]
         assert isinstance(raw, str) and raw.startswith("2026-09-14T12:00:00")
+
+    def test_status_present_with_zero_defaults(self):
+        # :166 TestStatusOmitEmpty — a freshly-specced VA still serializes a
+        # status object whose leaves carry zero values; flipping one bool
+        # keeps the key present
+        va = self._valid_va()
+        d = va.model_dump(by_alias=True, exclude_none=True, mode="json")
+        assert "status" in d
+        assert d["status"]["currentAlloc"]["accelerator"] in ("", "MI355X")
+        fresh = v1alpha1.VariantAutoscaling(
+            metadata=va.metadata, spec=va.spec
+        ).model_dump(by_alias=True, exclude_none=True, mode="json")
+        assert fresh["status"]["desiredOptimizedAlloc"]["numReplicas"] == 0
+        assert fresh["status"]["actuation"]["applied"] is False
+        va2 = v1alpha1.VariantAutoscaling(metadata=va.metadata, spec=va.spec)
+        va2.status.actuation.applied = True
+        d2 = va2.model_dump(by_alias=True, exclude_none=True, mode="json")
+        assert d2["status"]["actuation"]["applied"] is True
