@@ -356,6 +356,24 @@ class TestMultiVariant:
         # beyond-capacity sizing: unlimited mode has no cap
         assert results["c"]["vllm-c"] > 5
 
+    def test_multi_va_no_load_all_at_min_replicas(self, cluster, prom, registry):
+        """optimizer_test.go:245 — multiple VariantAutoscalings under zero
+        load all optimize to minNumReplicas (1) in a single global solve,
+        and every active VA receives an allocation."""
+        for name in ("a", "b", "c"):
+            make_deployment(cluster, name=f"idle-{name}")
+            make_va(cluster, name=f"idle-{name}", model_id="default/llama-8b")
+        set_load_metrics(prom, "default/llama-8b", "default", arrival_rps=0.0, out_tokens=0.0)
+        VariantAutoscalingReconciler(cluster, prom).reconcile()
+        vas = [
+            v
+            for v in cluster.list(type(get_va(cluster, "idle-a")))
+            if v.metadata.name.startswith("idle-")
+        ]
+        assert len(vas) == 3
+        for va in vas:
+            assert va.status.desired_optimized_alloc.num_replicas == 1, va.metadata.name
+
     def test_mixed_health_fleet(self, cluster, prom, registry):
         """One healthy VA + one with stale metrics + one deleted: only the
         healthy one is optimized, others untouched, cycle succeeds."""
