@@ -72,6 +72,27 @@ class TestCollectInventory:
         assert inv["MI355X"]["vendor"] == "amd.com"
         assert inv["H100"]["vendor"] == "nvidia.com"
 
+    def test_multiple_gpu_types_on_same_node(self):
+        # collector_test.go:182 — one node exposing two vendors' extended
+        # resources yields two inventory entries, each with nodes=1
+        client = InMemoryKubeClient()
+        n = Node(
+            metadata=ObjectMeta(
+                name="mixed", namespace="",
+                labels={
+                    "amd.com/gpu.product": "MI355X",
+                    "nvidia.com/gpu.product": "H100",
+                },
+            ),
+            status=NodeStatus(
+                allocatable={"amd.com/gpu": "8", "nvidia.com/gpu": "4"}
+            ),
+        )
+        client.create(n)
+        inv = collector.collect_inventory_k8s(client)
+        assert inv["MI355X"] == {"count": 8, "nodes": 1, "vendor": "amd.com"}
+        assert inv["H100"] == {"count": 4, "nodes": 1, "vendor": "nvidia.com"}
+
     def test_unparseable_count_skipped(self):
         client = InMemoryKubeClient()
         n = Node(
