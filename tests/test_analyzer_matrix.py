@@ -239,6 +239,37 @@ class TestConfigurationCheckTable:
             assert QueueAnalyzer(config, rs) is not None
 
 
+class TestNewQueueAnalyzerTable:
+    # queueanalyzer_test.go:26 TestNewQueueAnalyzer — constructor accepts
+    # every degenerate-but-valid request shape and rejects the invalid ones
+    @pytest.mark.parametrize(
+        "name,in_tok,out_tok,want_err",
+        [
+            ("no prefill", 0, 10, False),
+            ("no prefill, one output token", 0, 1, False),
+            ("no decode", 100, 1, False),
+            ("mixed prefill and decode", 200, 20, False),
+            ("zero input and output tokens", 0, 0, True),
+            ("negative tokens", -1, -1, True),
+            ("no decode, no first output token", 50, 0, True),
+        ],
+    )
+    def test_table(self, name, in_tok, out_tok, want_err):
+        from wva_amd.analyzer import AnalyzerError
+
+        config = Configuration(
+            max_batch_size=8, max_queue_size=16, service_parms=_valid_parms()
+        )
+        rs = RequestSize(avg_input_tokens=in_tok, avg_output_tokens=out_tok)
+        if want_err:
+            with pytest.raises(AnalyzerError):
+                QueueAnalyzer(config, rs)
+        else:
+            qa = QueueAnalyzer(config, rs)
+            # a constructible analyzer must expose a usable rate range
+            assert qa.rate_range.max > qa.rate_range.min > 0
+
+
 class TestRequestSizeCheckTable:
     # queueanalyzer_test.go:178 TestRequestSize_Check
     @pytest.mark.parametrize(
