@@ -332,6 +332,59 @@ class TestDecodeTimeTable:
         assert decode.decode_time(batch) == pytest.approx(expected)
 
 
+class TestAnalyzeRateSweepTable:
+    # queueanalyzer_test.go:357 TestQueueAnalyzer_Analyze — rate sweep
+    # across the operating range; every successful point returns sane
+    # metrics, zero/negative/beyond-max rates raise
+    def _qa(self):
+        config = Configuration(
+            max_batch_size=8, max_queue_size=16, service_parms=_valid_parms()
+        )
+        return QueueAnalyzer(
+            config, RequestSize(avg_input_tokens=100, avg_output_tokens=10)
+        )
+
+    @pytest.mark.parametrize(
+        "name,rate_of_range,want_err",
+        [
+            ("zero request rate", lambda r: 0.0, True),
+            ("negative request rate", lambda r: -1.0, True),
+            ("low request rate", lambda r: r.min * 0.5, False),
+            ("medium request rate", lambda r: (r.min + r.max) * 0.5, False),
+            ("high request rate within bounds", lambda r: r.max * 0.9, False),
+            ("request rate exceeding maximum", lambda r: r.max * 1.1, True),
+        ],
+    )
+    def test_table(self, name, rate_of_range, want_err):
+        from wva_amd.analyzer import AnalyzerError
+
+        qa = self._qa()
+        rate = rate_of_range(qa.rate_range)
+        if want_err:
+            with pytest.raises(AnalyzerError):
+                qa.analyze(rate)
+            return
+        m = qa.analyze(rate)
+        assert m.throughput >= 0
+        assert m.avg_resp_time >= 0
+        assert m.avg_wait_time >= 0
+        assert m.avg_num_in_serv >= 0
+        assert 0.0 <= m.rho <= 1.0
+        assert m.avg_prefill_time >= 0
+        assert m.avg_token_time >= 0
+
+    def test_metrics_monotone_in_rate(self):
+        # queueing sanity across the sweep: wait and occupancy grow with λ
+        qa = self._qa()
+        rates = [qa.rate_range.min * 0.5,
+                 (qa.rate_range.min + qa.rate_range.max) * 0.5,
+                 qa.rate_range.max * 0.9]
+        ms = [qa.analyze(r) for r in rates]
+        assert ms[0].avg_wait_time <= ms[1].avg_wait_time <= ms[2].avg_wait_time
+        assert ms[0].avg_num_in_serv <= ms[1].avg_num_in_serv <= ms[2].avg_num_in_serv
+        assert ms[0].throughput <= ms[1].throughput <= ms[2].throughput
+
+
 class TestMM1KUtilizationGrid:
     # queuemodel_test.go:152 TestMM1KModel_ProbabilityCalculation — the
     # utilization grid incl. the lambda == mu boundary
