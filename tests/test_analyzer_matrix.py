@@ -385,6 +385,26 @@ class TestAnalyzeRateSweepTable:
         assert ms[0].throughput <= ms[1].throughput <= ms[2].throughput
 
 
+class TestDiagnosticStrings:
+    # queueanalyzer_test.go:602 TestStringMethods — every analyzer type
+    # renders a non-empty diagnostic string naming its fields (dataclass
+    # reprs here; the Go side hand-writes String())
+    def test_reprs_name_fields(self):
+        parms = _valid_parms()
+        config = Configuration(
+            max_batch_size=8, max_queue_size=16, service_parms=parms
+        )
+        assert "max_batch_size" in repr(config)
+        assert "gamma" in repr(parms.prefill) and "delta" in repr(parms.prefill)
+        assert "alpha" in repr(parms.decode) and "beta" in repr(parms.decode)
+        rs = RequestSize(avg_input_tokens=100, avg_output_tokens=10)
+        assert "avg_input_tokens" in repr(rs)
+        qa = QueueAnalyzer(config, rs)
+        assert "min" in repr(qa.rate_range) and "max" in repr(qa.rate_range)
+        m = qa.analyze(qa.rate_range.max * 0.5)
+        assert "throughput" in repr(m) and "rho" in repr(m)
+
+
 class TestMM1KUtilizationGrid:
     # queuemodel_test.go:152 TestMM1KModel_ProbabilityCalculation — the
     # utilization grid incl. the lambda == mu boundary
