@@ -332,3 +332,76 @@ class TestSaturationPolicyEnum:
 
         for p in SaturationPolicy:
             assert SaturationPolicy.parse(str(p)) is p
+
+
+class TestServiceClassTargetTable:
+    """serviceclass_test.go:186-366 — model-target add/replace/remove/
+    update tables plus the spec round-trip (:368)."""
+
+    def _svc(self):
+        from wva_amd.core import ServiceClass
+
+        svc = ServiceClass("premium", 1)
+        svc.add_model_target(_mt("m1", itl=50.0, ttft=500.0))
+        return svc
+
+    def test_add_new_model_target(self):
+        svc = self._svc()
+        svc.add_model_target(_mt("m2", itl=80.0, ttft=2000.0))
+        assert svc.model_target("m2").itl == 80.0
+        assert svc.model_target("m1").itl == 50.0
+
+    def test_replace_existing_model_target(self):
+        # :203 — re-adding the same model overwrites its targets
+        svc = self._svc()
+        svc.add_model_target(_mt("m1", itl=25.0, ttft=250.0))
+        assert svc.model_target("m1").itl == 25.0
+        assert svc.model_target("m1").ttft == 250.0
+
+    def test_remove_existing_and_nonexistent(self):
+        # :240 — removing a missing target is a no-op
+        svc = self._svc()
+        svc.remove_model_target("m1")
+        assert svc.model_target("m1") is None
+        svc.remove_model_target("ghost")
+
+    def test_update_with_matching_name_and_priority(self):
+        from wva_amd.config import ServiceClassSpec
+
+        svc = self._svc()
+        ok = svc.update_model_targets(
+            ServiceClassSpec(name="premium", priority=1,
+                             model_targets=[_mt("m1", itl=10.0, ttft=100.0)])
+        )
+        assert ok and svc.model_target("m1").itl == 10.0
+
+    @pytest.mark.parametrize(
+        "name,priority",
+        [("other-class", 1), ("premium", 2)],
+    )
+    def test_update_with_mismatched_identity_rejected(self, name, priority):
+        # :324-341 — wrong name OR wrong priority leaves targets untouched
+        from wva_amd.config import ServiceClassSpec
+
+        svc = self._svc()
+        ok = svc.update_model_targets(
+            ServiceClassSpec(name=name, priority=priority,
+                             model_targets=[_mt("m1", itl=10.0, ttft=100.0)])
+        )
+        assert not ok and svc.model_target("m1").itl == 50.0
+
+    def test_spec_round_trip(self):
+        from wva_amd.core import ServiceClass
+
+        svc = self._svc()
+        svc.add_model_target(_mt("m2", itl=80.0, ttft=2000.0))
+        back = ServiceClass.from_spec(svc.spec())
+        assert back.name == svc.name and back.priority == svc.priority
+        assert back.model_target("m1").ttft == 500.0
+        assert back.model_target("m2").itl == 80.0
+
+
+def _mt(model, itl=0.0, ttft=0.0, tps=0.0):
+    from wva_amd.config import ModelTarget
+
+    return ModelTarget(model=model, slo_itl=itl, slo_ttft=ttft, slo_tps=tps)
