@@ -172,6 +172,28 @@ class TestSystem:
         with pytest.raises(KeyError):
             system.remove_accelerator("L40S")
 
+    def test_registry_removal_full_surface(self):
+        # system_test.go:570/923/1179 — model / service-class / capacity
+        # removal; missing model and class raise, missing capacity no-ops
+        system, _ = make_system()
+        model_name = next(iter(system.models))
+        system.remove_model(model_name)
+        assert system.model(model_name) is None
+        with pytest.raises(KeyError):
+            system.remove_model(model_name)
+
+        cls_name = next(iter(system.service_classes))
+        system.remove_service_class(cls_name)
+        assert system.service_class(cls_name) is None
+        with pytest.raises(KeyError):
+            system.remove_service_class(cls_name)
+
+        if system.capacity:
+            t = next(iter(system.capacity))
+            system.remove_capacity(t)
+            assert t not in system.capacity
+        system.remove_capacity("ghost-type")  # no-op
+
 
 class TestScaleAndReallocate:
     def test_scale_tracks_load_change(self):

@@ -95,8 +95,26 @@ class System:
             raise KeyError(f"server {name} not found")
         del self.servers[name]
 
+    def remove_model(self, name: str) -> None:
+        # system.go:570 TestSystem_RemoveModel — removing a missing model
+        # raises, matching remove_accelerator/remove_server semantics
+        if name not in self.models:
+            raise KeyError(f"model {name} not found")
+        del self.models[name]
+
+    def remove_service_class(self, name: str) -> None:
+        if name not in self.service_classes:
+            raise KeyError(f"service class {name} not found")
+        del self.service_classes[name]
+
     def set_capacity(self, cnt: AcceleratorCount) -> None:
         self.capacity[cnt.type] = cnt.count
+
+    def remove_capacity(self, type_name: str) -> None:
+        # system.go:1179 TestSystem_RemoveCapacity — no-op when absent
+        # (capacity is advisory; limited mode treats a missing type as
+        # unconstrained only if no pool was ever declared)
+        self.capacity.pop(type_name, None)
 
     # -- lookups ------------------------------------------------------------
     def accelerator(self, name: str) -> Optional[Accelerator]:
