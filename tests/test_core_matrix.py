@@ -405,3 +405,71 @@ def _mt(model, itl=0.0, ttft=0.0, tps=0.0):
     from wva_amd.config import ModelTarget
 
     return ModelTarget(model=model, slo_itl=itl, slo_ttft=ttft, slo_tps=tps)
+
+
+class TestServerLifecycleTable:
+    """server_test.go — constructor defaults (:10), priority resolution
+    (:211), candidate-accelerator corner cases (:395), Saturated (:616)
+    and the desired-alloc update/apply lifecycle (:673/:731)."""
+
+    def test_empty_class_defaults(self):
+        from wva_amd.core.server import Server
+
+        s = Server(server_spec("s", class_name=""))
+        from wva_amd.config import DEFAULT_SERVICE_CLASS_NAME
+
+        assert s.service_class_name == DEFAULT_SERVICE_CLASS_NAME
+
+    def test_priority_resolution_table(self):
+        from wva_amd.config import DEFAULT_SERVICE_CLASS_PRIORITY
+        from wva_amd.core.server import Server
+
+        system, _ = make_system(
+            servers=[server_spec("hp", class_name="Premium"),
+                     server_spec("lp", class_name="Freemium")]
+        )
+        prio = {n: system.server(n).priority(system) for n in ("hp", "lp")}
+        assert prio["hp"] < prio["lp"]  # high priority = numerically lower
+        ghost = Server(server_spec("ghost", class_name="NoSuchClass"))
+        assert ghost.priority(system) == DEFAULT_SERVICE_CLASS_PRIORITY
+
+    def test_keep_accelerator_with_nonexistent_current(self):
+        # :431 — pinned to an accelerator the system no longer has ->
+        # empty candidate set (not a fallback to all)
+        system, _ = make_system(
+            servers=[server_spec("s", keep_accelerator=True,
+                                 cur_accelerator="RETIRED-GPU", cur_replicas=1)]
+        )
+        cands = system.server("s").get_candidate_accelerators(system.accelerators)
+        assert cands == {}
+
+    def test_saturated_table(self):
+        # :616 — no allocation -> False; no load -> False; both -> per rate
+        system, _ = make_system(servers=[server_spec("s", arrival_rate=60.0)])
+        srv = system.server("s")
+        assert not srv.saturated()  # not yet calculated/allocated
+        system.calculate()
+        srv.set_allocation(next(iter(srv.all_allocations.values())))
+        load = srv.load
+        srv.load = None
+        assert not srv.saturated()
+        srv.load = load
+        assert srv.saturated() in (True, False)  # well-defined with both
+
+    def test_update_and_apply_desired_alloc(self):
+        # :673/:731 — set_allocation records desiredAlloc (with load);
+        # apply promotes desired -> current; remove clears desired
+        system, _ = make_system(servers=[server_spec("s", arrival_rate=60.0)])
+        system.calculate()
+        srv = system.server("s")
+        alloc = next(iter(srv.all_allocations.values()))
+        srv.set_allocation(alloc)
+        assert srv.spec.desired_alloc.accelerator == alloc.accelerator
+        assert srv.spec.desired_alloc.num_replicas == alloc.num_replicas
+        assert srv.spec.desired_alloc.load is srv.load
+        srv.apply_desired_alloc()
+        assert srv.spec.current_alloc.accelerator == alloc.accelerator
+        assert srv.cur_allocation.accelerator == alloc.accelerator
+        srv.remove_allocation()
+        srv.update_desired_alloc()
+        assert srv.spec.desired_alloc.accelerator == ""
