@@ -1,4 +1,6 @@
-"""Kubernetes API error classes (apimachinery apierrors analog)."""
+"""Kubernetes API error classes — the apimachinery `apierrors` analog
+(the reference relies on client-go's IsConflict/IsNotFound/IsGone
+predicates inside its retry helpers, internal/utils/utils.go:58-104)."""
 
 
 class KubeError(Exception):

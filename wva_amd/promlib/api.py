@@ -1,4 +1,9 @@
-"""PromAPI adapter over the promlib store — what the collector talks to."""
+"""PromAPI adapter over the promlib store — what the collector talks to.
+
+No reference counterpart: the reference's e2e tier needs a live
+kube-prometheus-stack (test/utils deploy-llm-d.sh installs one with
+self-signed TLS); promlib replaces that dependency in-process so the
+same collector queries run offline."""
 
 from __future__ import annotations
 

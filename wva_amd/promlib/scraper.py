@@ -1,4 +1,9 @@
-"""Target scraper: Prometheus text-format -> TimeSeriesStore."""
+"""Target scraper: Prometheus text-format -> TimeSeriesStore.
+
+Stands in for the Prometheus server's scrape loop (no reference
+counterpart — the reference assumes a cluster Prometheus); adds the
+per-target `instance` label exactly as a real scraper would, which is
+what keeps multi-replica emulator fleets' series distinct."""
 
 from __future__ import annotations
 
