@@ -259,6 +259,40 @@ class TestSLOChangeMidRun:
         assert get_va(cluster).status.desired_optimized_alloc.num_replicas == before
 
 
+class TestCostChangeMidRun:
+    def test_unit_cost_update_reprices_next_cycle(self, cluster, prom, registry):
+        """The accelerator unit-cost ConfigMap is also re-read per cycle
+        (variantautoscaling_controller.go:108): repricing MI355X changes
+        the scraped currentAlloc.variantCost on the next reconcile."""
+        import json
+
+        from wva_amd.controller.reconciler import (
+            ACCELERATOR_COSTS_CM,
+            CONFIG_MAP_NAMESPACE,
+        )
+        from wva_amd.kube import ConfigMap
+
+        make_deployment(cluster, replicas=2)
+        make_va(cluster)
+        set_load_metrics(prom, "default/llama-8b", "default", arrival_rps=2.0)
+        rec = VariantAutoscalingReconciler(cluster, prom)
+        rec.reconcile()
+        assert float(get_va(cluster).status.current_alloc.variant_cost) == pytest.approx(
+            170.0  # 2 replicas x 85.00
+        )
+
+        cm = cluster.get(ConfigMap, ACCELERATOR_COSTS_CM, CONFIG_MAP_NAMESPACE)
+        cm.data = dict(cm.data)
+        entry = json.loads(cm.data["MI355X"])
+        entry["cost"] = "100.00"
+        cm.data["MI355X"] = json.dumps(entry)
+        cluster.update(cm)
+        rec.reconcile()
+        assert float(get_va(cluster).status.current_alloc.variant_cost) == pytest.approx(
+            200.0
+        )
+
+
 class TestOwnerRefGC:
     def test_va_garbage_collected_on_deployment_delete(self, cluster, prom, registry):
         make_deployment(cluster)
