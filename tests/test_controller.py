@@ -259,6 +259,31 @@ class TestSLOChangeMidRun:
         assert get_va(cluster).status.desired_optimized_alloc.num_replicas == before
 
 
+class TestPerfParmsUpdateMidRun:
+    def test_refitted_perf_parms_resize_next_cycle(self, cluster, prom, registry):
+        """Editing a VA's spec.modelProfile perfParms (e.g. after a
+        re-fit on new firmware) re-sizes on the next reconcile — the
+        system spec is rebuilt from live VAs every cycle."""
+        make_deployment(cluster, replicas=1)
+        make_va(cluster)  # alpha=6.958, beta=0.042
+        set_load_metrics(
+            prom, "default/llama-8b", "default", arrival_rps=14.0, out_tokens=200.0
+        )
+        rec = VariantAutoscalingReconciler(cluster, prom)
+        rec.reconcile()
+        before = get_va(cluster).status.desired_optimized_alloc.num_replicas
+
+        va = get_va(cluster)
+        # a 3x slower decode law (regressed kernel, say) at the same load
+        va.spec.model_profile.accelerators[0].perf_parms.decode_parms = {
+            "alpha": "20.874", "beta": "0.126"
+        }
+        cluster.update(va)
+        rec.reconcile()
+        after = get_va(cluster).status.desired_optimized_alloc.num_replicas
+        assert after > before
+
+
 class TestCostChangeMidRun:
     def test_unit_cost_update_reprices_next_cycle(self, cluster, prom, registry):
         """The accelerator unit-cost ConfigMap is also re-read per cycle
