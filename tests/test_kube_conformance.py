@@ -591,3 +591,36 @@ class TestEventLogStateMachine:
             # invariant: list matches the live set
             names = {d.metadata.name for d in store.list(Deployment, "ns")}
             assert names == live
+
+
+class TestErrorRetryability:
+    """Retry-gate semantics (internal/utils/utils.go:58-104 — client-go's
+    IsNotFound/IsInvalid/IsForbidden short-circuit the backoff loops;
+    conflicts retry with a fresh read; 410 forces a re-list)."""
+
+    def test_flags(self):
+        from wva_amd.kube.errors import (
+            ConflictError,
+            ForbiddenError,
+            InvalidError,
+            KubeError,
+            NotFoundError,
+        )
+
+        assert KubeError("x").retryable is True
+        assert ConflictError("x").retryable is True
+        for cls in (NotFoundError, InvalidError, ForbiddenError, GoneError):
+            assert cls("x").retryable is False, cls.__name__
+
+    def test_backoff_short_circuits_on_non_retryable(self):
+        # a NotFound must surface immediately, not after the backoff budget
+        import time
+
+        from wva_amd.controller.utils import get_deployment_with_backoff
+        from wva_amd.kube import InMemoryKubeClient, NotFoundError
+
+        client = InMemoryKubeClient()
+        t0 = time.perf_counter()
+        with pytest.raises(NotFoundError):
+            get_deployment_with_backoff(client, "ghost", "default")
+        assert time.perf_counter() - t0 < 0.5  # no 100ms*2^5 backoff spent
