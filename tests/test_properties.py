@@ -257,3 +257,45 @@ class TestMG1Properties:
         w_hi = qa_hi.analyze(rate).avg_wait_time
         assert w_lo <= w_hi + 1e-12
         assert qa_1.analyze(rate).avg_wait_time == qa_ref.analyze(rate).avg_wait_time
+
+
+class TestPromlibProperties:
+    """Randomized cross-check of the PromQL-subset rate() against a
+    direct computation on the raw samples (counter semantics incl.
+    resets), for any monotone-with-resets sample path."""
+
+    @given(
+        increments=st.lists(st.floats(0.0, 50.0), min_size=3, max_size=40),
+        reset_at=st.integers(0, 39),
+        step_s=st.floats(0.5, 10.0),
+    )
+    @settings(max_examples=60, deadline=None)
+    def test_rate_matches_manual(self, increments, reset_at, step_s):
+        from wva_amd.promlib.promql import evaluate
+        from wva_amd.promlib.store import TimeSeriesStore
+
+        store = TimeSeriesStore()
+        t0 = 1_000_000.0
+        labels = {"__name__": "ctr_total", "job": "x"}
+        value = 0.0
+        samples = []
+        for i, inc in enumerate(increments):
+            if i == reset_at and i > 0:
+                value = 0.0  # counter reset
+            value += inc
+            t = t0 + i * step_s
+            store.add_sample("ctr_total", labels, value, ts=t)
+            samples.append((t, value))
+        now = t0 + (len(increments) - 1) * step_s
+        window = now - t0 + 1e-9
+        out = evaluate(f"rate(ctr_total[{int(window) + 1}s])", store, now=now)
+        # manual: sum of positive deltas (resets add the post-reset value)
+        in_window = [s for s in samples if s[0] >= now - (int(window) + 1)]
+        if len(in_window) < 2:
+            return
+        total = 0.0
+        for (t_a, v_a), (t_b, v_b) in zip(in_window, in_window[1:]):
+            total += (v_b - v_a) if v_b >= v_a else v_b
+        expected = total / (in_window[-1][0] - in_window[0][0])
+        assert len(out) == 1
+        assert out[0].value == pytest.approx(expected, rel=1e-9, abs=1e-12)

@@ -11,8 +11,9 @@ wva_amd/controller/collector.py and the KEDA/HPA queries):
 
 Semantics follow Prometheus: instant selectors return the latest sample
 per series within a 5-minute lookback (with its original timestamp);
-rate() needs >=2 samples in the window and divides the value delta by the
-time delta (counter resets clamp to 0); aggregations collapse matching
+rate() needs >=2 samples in the window and divides the reset-corrected
+sum of sample deltas by the time delta (a mid-window counter reset
+contributes the post-reset value, Prometheus-style); aggregations collapse matching
 series; '/' divides two single-sample vectors and returns empty when
 either side is empty.
 """
@@ -163,12 +164,19 @@ def _eval_node(node, store: TimeSeriesStore, now: float) -> List[Sample]:
             pts = series.range(now - window, now)
             if len(pts) < 2:
                 continue
-            (t0, v0), (t1, v1) = pts[0], pts[-1]
+            t0, t1 = pts[0][0], pts[-1][0]
             if t1 <= t0:
                 continue
-            delta = v1 - v0
-            if delta < 0:  # counter reset
-                delta = v1
+            # per-adjacent-pair counter-reset correction, as Prometheus's
+            # extrapolatedRate does: a mid-window reset contributes the
+            # post-reset value instead of poisoning the endpoint delta
+            # (matters when an emulator/vLLM pod restarts inside the
+            # rate window during fleet resizes)
+            delta = 0.0
+            prev = pts[0][1]
+            for _, v in pts[1:]:
+                delta += (v - prev) if v >= prev else v
+                prev = v
             out.append(Sample(value=delta / (t1 - t0), timestamp=now, labels=dict(series.labels)))
         return out
     if kind == "agg":
