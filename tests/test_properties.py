@@ -411,3 +411,37 @@ class TestSchemaValidatorProperties:
         obj["spec"]["modelProfile"]["accelerators"][0][field] = value
         with pytest.raises(SchemaValidationError):
             CRDValidator().validate(obj)
+
+
+class TestChunkedListProperties:
+    """Pagination fuzz on the in-memory store (the same contract the stub
+    apiserver serves over HTTP): for any object count and page size the
+    chunks partition the collection — no duplicates, no gaps, stable
+    order, and every chunk reports the resourceVersion the list started
+    at."""
+
+    @given(n=st.integers(0, 57), limit=st.integers(1, 19))
+    @settings(max_examples=40, deadline=None)
+    def test_chunks_partition_collection(self, n, limit):
+        from wva_amd.api.v1alpha1.types import ObjectMeta
+        from wva_amd.kube import Deployment, InMemoryKubeClient
+
+        client = InMemoryKubeClient()
+        for i in range(n):
+            client.create(
+                Deployment(metadata=ObjectMeta(name=f"d{i:03d}", namespace="ns"))
+            )
+        seen = []
+        rvs = set()
+        token = ""
+        while True:
+            items, rv, token = client.list_meta(
+                Deployment, "ns", limit=limit, continue_token=token
+            )
+            seen.extend(o.metadata.name for o in items)
+            rvs.add(rv)
+            if not token:
+                break
+            assert len(items) == limit  # only the last chunk may be short
+        assert seen == sorted(f"d{i:03d}" for i in range(n))
+        assert len(rvs) <= 1  # one logical list -> one resourceVersion
