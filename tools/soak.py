@@ -43,6 +43,7 @@ def run_soak(
     max_replicas: int = 6,
     variants: int = 1,
     quiet: bool = False,
+    scale_down_stabilization_s: float = 0.0,
 ) -> dict:
     """Run the closed loop; returns the result dict (importable from
     bench.py so BENCH records carry measured attainment)."""
@@ -123,6 +124,7 @@ def run_soak(
     import contextlib
 
     trajectory = []
+    recommendations = []  # (wall_time, desired) for scale-down stabilization
     with contextlib.ExitStack() as stack:
         fleet = stack.enter_context(EmulatorFleet(settings, max_replicas=args.max_replicas))
         for url in fleet.urls:
@@ -166,8 +168,23 @@ def run_soak(
                 desired = va.status.desired_optimized_alloc.num_replicas
                 for name in names:
                     simulate_hpa(cluster, registry, name=name)
-                # actuate: the simulated HPA resizes the serving fleet
-                fleet.scale(max(desired, 1))
+                # actuate: the simulated HPA resizes the serving fleet.
+                # With scale_down_stabilization_s > 0 it applies HPA's
+                # scaleDown stabilizationWindowSeconds semantics — the
+                # applied size is the MAX recommendation in the trailing
+                # window (scale-up stays instant), mirroring the shipped
+                # deploy/integrations/hpa.yaml (120 s)
+                now_w = time.time()
+                recommendations.append((now_w, desired))
+                if scale_down_stabilization_s > 0:
+                    cutoff = now_w - scale_down_stabilization_s
+                    recommendations[:] = [
+                        (t, d) for t, d in recommendations if t >= cutoff
+                    ] or [(now_w, desired)]
+                    applied = max(d for _, d in recommendations)
+                else:
+                    applied = desired
+                fleet.scale(max(applied, 1))
 
                 arrival_rpm = float(va.status.current_alloc.load.arrival_rate)
                 per_replica = (arrival_rpm / 60.0) / max(desired, 1)
@@ -231,6 +248,8 @@ def main() -> None:
     ap.add_argument("--stage-seconds", type=float, default=8.0)
     ap.add_argument("--max-replicas", type=int, default=6, help="fleet instances pre-started")
     ap.add_argument("--variants", type=int, default=1, help="number of variants (extras carry steady load)")
+    ap.add_argument("--scale-down-stabilization", type=float, default=0.0,
+                    help="HPA scaleDown stabilizationWindowSeconds to emulate (0 = apply instantly)")
     ap.add_argument("--out", default="")
     args = ap.parse_args()
     result = run_soak(
@@ -238,6 +257,7 @@ def main() -> None:
         stage_seconds=args.stage_seconds,
         max_replicas=args.max_replicas,
         variants=args.variants,
+        scale_down_stabilization_s=args.scale_down_stabilization,
     )
     if args.out:
         with open(args.out, "w") as f:
