@@ -225,3 +225,42 @@ class TestSloObserver:
         assert p is not None
         assert p.itl_ms > 12.0  # at least alpha
         assert p.ttft_ms > 0
+
+
+class TestSoakStabilizationWindow:
+    def test_scale_down_held_within_window(self):
+        """HPA scaleDown stabilization emulation: with a window far longer
+        than the run, the applied fleet size never decreases even when the
+        sizing recommendation drops (scale-up stays instant)."""
+        import sys
+
+        sys.path.insert(0, "tools")
+        from soak import run_soak
+
+        r = run_soak(
+            stages=(2.0, 4.0, 6.0, 2.0),
+            stage_seconds=3.0,
+            max_replicas=6,
+            quiet=True,
+            scale_down_stabilization_s=3600.0,
+        )
+        fleet_sizes = [e["fleet_replicas"] for e in r["trajectory"]]
+        assert all(b >= a for a, b in zip(fleet_sizes, fleet_sizes[1:])), fleet_sizes
+
+    def test_zero_window_applies_instantly(self):
+        import sys
+
+        sys.path.insert(0, "tools")
+        from soak import run_soak
+
+        r = run_soak(
+            stages=(6.0, 2.0, 2.0),
+            stage_seconds=3.0,
+            max_replicas=6,
+            quiet=True,
+        )
+        traj = r["trajectory"]
+        # with no window the fleet tracks the recommendation exactly
+        assert all(
+            e["fleet_replicas"] == max(e["desired_replicas"], 1) for e in traj
+        )
