@@ -134,8 +134,17 @@ class TestIntegrations:
     def test_emulator_manifest(self):
         docs = load("emulator/vllm-emulator.yaml")
         deploy = docs[0]
-        env = {e["name"]: e.get("value") for e in deploy["spec"]["template"]["spec"]["containers"][0]["env"]}
+        container = deploy["spec"]["template"]["spec"]["containers"][0]
+        env = {e["name"]: e.get("value") for e in container["env"]}
         assert env["MEM_SIZE"] == "294912"  # 288 GB MI355X
+        # e2e_test.go:293 — the emulator pod requests the (emulated) GPU
+        # extended resource so capacity accounting sees a real GPU pod
+        assert container["resources"]["limits"]["amd.com/gpu"] == "1"
+        assert container["resources"]["requests"]["amd.com/gpu"] == "1"
+        # service selector matches the pod labels (e2e_test.go:261)
+        svc = next(d for d in docs if d and d.get("kind") == "Service")
+        pod_labels = deploy["spec"]["template"]["metadata"]["labels"]
+        assert all(pod_labels.get(k) == v for k, v in svc["spec"]["selector"].items())
 
 
 class TestMainEntry:
